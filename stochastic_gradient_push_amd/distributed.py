@@ -1,0 +1,588 @@
+"""GossipDataParallel — synchronous / overlap SGP and D-PSGD model wrapper.
+
+Behavioral parity with reference gossip/distributed.py:39-589, rebuilt
+MI355X-first:
+
+* **One process per GPU** is the native mode (`torch.distributed` over
+  RCCL/xGMI).  The reference's single-process multi-GPU replica tier
+  (DataParallel-style replicate/scatter/gather, reference
+  distributed.py:91-99, 231-254) is intentionally not reproduced — it is
+  a PyTorch-1.x idiom superseded by process-per-GPU; the multi-process
+  intra-node tier (``nprocs_per_node > 1``: params broadcast from the
+  local master, grads all-reduced on a per-node group, gossip done by the
+  local master only, reference distributed.py:62-78, 278-296, 551-562) IS
+  supported.
+* **Flat parameter buffer**: all trainable params live as views of one
+  contiguous device buffer (:class:`~..ops.flat.FlatParams`), so every
+  push-sum state transition (bias, de-bias, residual-add, pack) is a
+  single fused kernel launch over ~100 MB instead of ~161 per-tensor ops
+  (reference distributed.py:298-314, 372-379, 402-425).
+* **True p2p gossip** on a dedicated process group (its own RCCL
+  communicator), so the background gossip thread's sends/receives never
+  contend with main-thread collectives.  See
+  :mod:`~stochastic_gradient_push_amd.gossiper`.
+* The train<->gossip concurrency protocol is the reference's proven
+  Event ping-pong (gossip_flag/train_flag strict buffer ownership,
+  heartbeat timeout, interrupted-gossip ``ps_weight == -1`` sentinel,
+  reference distributed.py:336-387, 459-510), with gossip work enqueued
+  on a side HIP stream.
+
+Push-sum numerics note (load-bearing, reference distributed.py:565): the
+optimizer steps on the *biased* numerator parameters; the backward hook
+re-biases before the step.  ``lazy_mixing`` (regular mixing, no overlap,
+no async) folds the bias/de-bias scalings into the residual-merge so the
+parameters never need rescaling at all.
+"""
+
+import functools
+import threading
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+from torch.autograd import Variable
+from torch.nn.modules import Module
+
+from . import ops
+from .gossiper import PushPull, PushSum
+from .graphs import NPeerDynamicDirectedExponentialGraph as NPDDEGraph
+from .mixing import UniformMixing
+from .ops.flat import FlatParams
+from .utils.helpers import (
+    communicate,
+    create_process_group,
+    make_logger,
+)
+
+HEARTBEAT_TIMEOUT = 300  # max seconds to wait on the gossip thread
+
+
+class GossipDataParallel(Module):
+    """Distributed gossip model wrapper (API parity: reference
+    distributed.py:42-45)."""
+
+    def __init__(
+        self,
+        module: Module,
+        device_ids=None,
+        rank=None,
+        world_size=None,
+        graph=None,
+        mixing=None,
+        comm_device=None,
+        push_sum: bool = True,
+        overlap: bool = False,
+        synch_freq: int = 0,
+        verbose: bool = False,
+        use_streams: bool = True,
+        nprocs_per_node: int = 1,
+        local_node_group=None,
+        flatten_grads: bool = True,
+    ):
+        super().__init__()
+
+        if world_size is None or rank is None:
+            assert dist.is_initialized()
+            rank = dist.get_rank()
+            world_size = dist.get_world_size()
+        self.process_rank = rank
+        self.nprocs_per_node = nprocs_per_node
+
+        # node-level rank/world (reference distributed.py:62-66)
+        if nprocs_per_node > 1:
+            self.local_rank = self.process_rank % nprocs_per_node
+            world_size //= nprocs_per_node
+            rank //= nprocs_per_node
+            if local_node_group is None:
+                for node in range(world_size):
+                    node_ranks = list(
+                        range(node * nprocs_per_node, (node + 1) * nprocs_per_node)
+                    )
+                    grp = create_process_group(node_ranks)
+                    if self.process_rank in node_ranks:
+                        self.local_node_group = grp
+            else:
+                self.local_node_group = local_node_group
+        else:
+            self.local_rank = 0
+            self.local_node_group = None
+
+        self.module = module
+        first_param_dtype = next(module.parameters()).dtype
+        self._cuda = next(module.parameters()).is_cuda
+
+        # communication device (reference distributed.py:101-105)
+        if comm_device is None:
+            cpu_comm = dist.get_backend() == "gloo"
+            comm_device = torch.device("cpu") if cpu_comm else torch.device("cuda")
+        self.__cpu_comm = comm_device.type == "cpu"
+
+        self.distributed = world_size > 1
+        if graph is None and self.distributed:
+            graph = NPDDEGraph(rank, world_size, nprocs_per_node, self.local_rank)
+        if mixing is None and self.distributed:
+            mixing = UniformMixing(graph, comm_device)
+
+        self.dist_config = {
+            "verbose": verbose,
+            "comm_device": comm_device,
+            "graph": graph,
+            "mixing": mixing,
+            "push_sum": push_sum,
+            "rank": rank,
+            "process_rank": self.process_rank,
+            "world_size": world_size,
+            "cpu_comm": self.__cpu_comm,
+            "gossipers": {},
+        }
+        self.overlap = overlap
+        self.synch_freq = synch_freq
+        self.num_updates = 0
+        self.asynch = synch_freq > 0
+        self.logger = make_logger(rank, verbose)
+
+        # flat parameter/grad views — the central data structure
+        self.flatp = FlatParams(module, flatten_grads=flatten_grads)
+        self._flatten_grads = flatten_grads
+
+        # push-sum state
+        self.ps_weight = torch.ones(1, device=comm_device, dtype=first_param_dtype)
+        self.is_ps_numerator = False
+        self.nprocs_per_node_device = torch.tensor(
+            [nprocs_per_node], device=comm_device, dtype=first_param_dtype
+        )
+
+        self.gossip_enable = True
+        self.gossiping = False
+        self.params_mixed = True
+        self.gossip_ps_factor = torch.zeros(
+            1, device=comm_device, dtype=first_param_dtype
+        )
+        self.gossip_ps_weight = self.ps_weight.clone()
+
+        if not self.distributed:
+            # single-process job: nothing to gossip; wrapper is transparent
+            self.lazy_mixing = True
+            self.lazy_ps_factor = self.gossip_ps_factor.clone()
+            self.gossip_thread = None
+            return
+
+        # staging + comm buffers (flat; reference used per-tensor clones,
+        # distributed.py:149-155)
+        self.gossip_device_buffer = torch.empty_like(self.flatp.flat)
+        if self.__cpu_comm:
+            staged = self.flatp.flat.detach().cpu().clone()
+            if torch.cuda.is_available():
+                staged = staged.pin_memory()
+            self.gossip_params = staged
+        else:
+            # device comm: the staging buffer IS the comm buffer
+            self.gossip_params = self.gossip_device_buffer
+
+        # dedicated process group => separate RCCL communicator for the
+        # gossip thread (thread isolation; see module docstring)
+        self.gossip_group = create_process_group(
+            list(range(dist.get_world_size()))
+        )
+
+        # control objects (reference distributed.py:157-165)
+        self.gossip_lock = threading.Lock()
+        self.gossip_flag = threading.Event()
+        self.train_flag = threading.Event()
+        if self._cuda and not self.__cpu_comm and use_streams:
+            self.gossip_stream = torch.cuda.Stream()
+        elif self._cuda:
+            self.gossip_stream = torch.cuda.current_stream()
+        else:
+            self.gossip_stream = None
+
+        if self.process_rank % nprocs_per_node == 0:
+            self.gossip_thread = threading.Thread(
+                target=GossipDataParallel._gossip_target,
+                args=(
+                    self.dist_config,
+                    self.gossip_flag,
+                    self.train_flag,
+                    self.gossip_lock,
+                    self.gossip_params,
+                    self.gossip_device_buffer,
+                    self.gossip_ps_weight,
+                    self.gossip_ps_factor,
+                    self.gossip_stream,
+                    self.gossip_group,
+                ),
+                daemon=True,
+                name="Gossip-Thread",
+            )
+            self.gossip_thread.start()
+        else:
+            self.gossip_thread = None
+            self.gossip_flag.set()
+
+        # wait for gossip thread to finish initialization
+        self.gossip_flag.wait()
+        self.gossip_flag.clear()
+
+        # lazy mixing decision (reference distributed.py:188-191)
+        self.lazy_mixing = (
+            not self.asynch
+            and self.dist_config["mixing"].is_regular()
+            and not self.overlap
+        )
+        self.lazy_ps_factor = self.gossip_ps_factor.clone()
+        self.logger.debug(f"lazy mixing: {self.lazy_mixing}")
+
+        self.__register_hooks()
+
+    # -- public API ---------------------------------------------------------
+
+    def update_gossiper(self, attr, val):
+        """Thread-safe attribute update on the live gossipers (reference
+        distributed.py:197-207)."""
+        if not self.distributed:
+            return
+        with self.gossip_lock:
+            for gossiper in self.dist_config["gossipers"].values():
+                if val == getattr(gossiper, attr):
+                    continue
+                setattr(gossiper, attr, val)
+
+    def state_dict(self, finish_gossip: bool = True):
+        """Checkpoint wrapper embedding push-sum state; drains in-flight
+        gossip first so peer-sent mass is not lost (reference
+        distributed.py:209-222)."""
+        if finish_gossip:
+            self._query_gossip_queue()
+        return {
+            "state_dict": super().state_dict(),
+            "ps_weight": self.ps_weight.cpu(),
+            "is_ps_numerator": self.is_ps_numerator,
+        }
+
+    def load_state_dict(self, load_dict):
+        super().load_state_dict(load_dict["state_dict"])
+        self.ps_weight = load_dict["ps_weight"].to(
+            device=self.dist_config["comm_device"]
+        )
+        self.is_ps_numerator = load_dict["is_ps_numerator"]
+        # loading re-pointed nothing: params are views, load_state_dict
+        # copies in place, so the flat buffer is already current
+
+    def forward(self, *inputs, **kwargs):
+        if self.distributed and self.nprocs_per_node > 1:
+            self._sync_params_multiprocess()
+        return self.module(*inputs, **kwargs)
+
+    def train(self, mode: bool = True):
+        super().train(mode)
+        self.gossip_enable = self.distributed and mode
+        return self
+
+    def eval(self):
+        super().eval()
+        self.gossip_enable = False
+        if self.distributed:
+            self._query_gossip_queue(non_blocking=self.asynch)
+        return self
+
+    def block(self):
+        if not self.distributed:
+            return
+        self.logger.info("blocking")
+        dist.barrier()
+
+    def sync_comms(self):
+        self._query_gossip_queue(non_blocking=False)
+
+    # -- intra-node multiprocess tier ---------------------------------------
+
+    def _sync_params_multiprocess(self):
+        """Broadcast params+buffers from the local master (reference
+        distributed.py:278-296); the flat param broadcast is one op."""
+        src = self.dist_config["rank"] * self.nprocs_per_node
+        dist.broadcast(self.flatp.flat, src=src, group=self.local_node_group)
+        buffers = [b.data for b in self.module.buffers()]
+        if buffers:
+            communicate(
+                buffers,
+                functools.partial(
+                    dist.broadcast, src=src, group=self.local_node_group
+                ),
+            )
+
+    # -- push-sum state machine ---------------------------------------------
+
+    def ps_numerator(self):
+        """params *= ps_weight (reference distributed.py:298-305)."""
+        if not self.is_ps_numerator:
+            if not self.lazy_mixing:
+                ops.scale_(
+                    self.flatp.flat,
+                    self.ps_weight.to(self.flatp.flat.dtype),
+                )
+            self.is_ps_numerator = True
+
+    def unbias(self):
+        """params /= ps_weight (reference distributed.py:307-314)."""
+        if self.is_ps_numerator:
+            if not self.lazy_mixing:
+                ops.scale_(
+                    self.flatp.flat,
+                    (1.0 / self.ps_weight).to(self.flatp.flat.dtype),
+                )
+            self.is_ps_numerator = False
+
+    def _query_gossip_queue(self, non_blocking: bool = False):
+        """Merge received push-sum residuals into the model (reference
+        distributed.py:336-387)."""
+        if not self.distributed or not self.gossip_enable:
+            return False
+
+        if not self.gossiping:
+            if self.process_rank % self.nprocs_per_node == 0:
+                self.logger.warning("not gossiping right now")
+            return False
+
+        if not non_blocking:
+            if not self.gossip_flag.wait(timeout=HEARTBEAT_TIMEOUT):
+                raise RuntimeError("Gossip flag timeout")  # heartbeat monitor
+
+        if self.gossip_flag.is_set():
+            # interrupted gossip — re-arm and retry next iteration
+            # (reference distributed.py:359-364)
+            if self.gossip_ps_weight[0] == -1:
+                self.gossip_flag.clear()
+                self.params_mixed = True
+                self.gossiping = False
+                self.transfer_params(mix=False)
+                return False
+
+            self.lazy_ps_factor.copy_(self.gossip_ps_factor)
+            self.ps_numerator()
+            self.ps_weight += self.gossip_ps_weight
+            if self.lazy_mixing:
+                self.ps_weight *= self.lazy_ps_factor
+                ops.add_scale_(
+                    self.flatp.flat,
+                    self.gossip_device_buffer,
+                    self.lazy_ps_factor.to(self.flatp.flat.dtype),
+                )
+            else:
+                ops.add_scale_(self.flatp.flat, self.gossip_device_buffer, 1.0)
+
+            self.logger.debug(f"updated ps-weight {self.ps_weight}")
+            self.gossip_flag.clear()
+            self.params_mixed = True
+            self.gossiping = False
+            return True
+        return False
+
+    def transfer_params(self, mix: bool = True) -> bool:
+        """Pack a copy of the (pre-scaled) params for the gossip thread and
+        wake it (reference distributed.py:389-434)."""
+        if (
+            not self.distributed
+            or not self.gossip_enable
+            or self.process_rank % self.nprocs_per_node != 0
+        ):
+            return False
+        if not self.params_mixed:
+            self.logger.warning("params not mixed")
+            return False
+
+        mix = mix and not self.lazy_mixing
+
+        self.ps_numerator()
+        if mix:
+            self.ps_weight *= self.gossip_ps_factor
+        self.gossip_ps_weight.copy_(self.ps_weight)
+
+        # fused: params *= factor (if mixing) and pack into staging buffer
+        # in one pass (reference distributed.py:409-418 did ~161 mul_ +
+        # ~161 copy_)
+        if mix:
+            ops.pack_mix_(
+                self.flatp.flat,
+                self.gossip_device_buffer,
+                self.gossip_ps_factor.to(self.flatp.flat.dtype),
+            )
+        else:
+            ops.pack_mix_(self.flatp.flat, self.gossip_device_buffer, 1.0)
+
+        if self._cuda:
+            # hand the staging buffer to the gossip stream; async copy to
+            # pinned host memory only in CPU-comm mode
+            self.gossip_stream.wait_stream(torch.cuda.current_stream())
+            if self.__cpu_comm:
+                with torch.cuda.stream(self.gossip_stream):
+                    self.gossip_params.copy_(
+                        self.gossip_device_buffer, non_blocking=True
+                    )
+        elif self.__cpu_comm and self.gossip_params.data_ptr() != \
+                self.gossip_device_buffer.data_ptr():
+            self.gossip_params.copy_(self.gossip_device_buffer)
+
+        self.params_mixed = False
+        self.gossiping = True
+        self.train_flag.set()
+        return True
+
+    # -- gossip thread -------------------------------------------------------
+
+    @staticmethod
+    def _gossip_into_receive_buffer(
+        send_buffer, gossiper, receive_buffer, gossip_ps_weight,
+        gossip_lock, dist_config,
+    ):
+        with gossip_lock:
+            in_msg, ps_weight = gossiper.mix(
+                send_buffer, gossip_ps_weight, residual=True
+            )
+            ps_factor = gossiper.mixing_weights["lo"]
+        # deposit the received residual for the train thread
+        if receive_buffer.data_ptr() != in_msg.data_ptr():
+            receive_buffer.copy_(
+                in_msg, non_blocking=dist_config["cpu_comm"]
+            )
+        return ps_weight, ps_factor
+
+    @staticmethod
+    def _gossip_target(
+        dist_config, gossip_flag, train_flag, gossip_lock, gossip_params,
+        gossip_device_buffer, gossip_ps_weight, gossip_ps_factor,
+        gossip_stream, gossip_group,
+    ):
+        """Background gossip loop (reference distributed.py:459-510)."""
+        logger = make_logger(dist_config["rank"], dist_config["verbose"])
+
+        gossiper_class = PushSum if dist_config["push_sum"] else PushPull
+        gossiper = gossiper_class(
+            gossip_params,
+            device=dist_config["comm_device"],
+            graph=dist_config["graph"],
+            mixing=dist_config["mixing"],
+            rank=dist_config["process_rank"],
+            world_size=dist_config["world_size"],
+            logger=logger,
+            group=gossip_group,
+        )
+        dist_config["gossipers"] = {gossip_params.dtype: gossiper}
+        gossip_ps_factor.data.copy_(gossiper.mixing_weights["lo"])
+        gossip_flag.set()
+
+        while True:
+            train_flag.wait()
+            logger.debug("received train-flag")
+            try:
+                if gossip_stream is not None:
+                    with torch.cuda.stream(gossip_stream):
+                        ps_weight, ps_factor = (
+                            GossipDataParallel._gossip_into_receive_buffer(
+                                gossip_params, gossiper, gossip_device_buffer,
+                                gossip_ps_weight, gossip_lock, dist_config,
+                            )
+                        )
+                else:
+                    ps_weight, ps_factor = (
+                        GossipDataParallel._gossip_into_receive_buffer(
+                            gossip_params, gossiper, gossip_device_buffer,
+                            gossip_ps_weight, gossip_lock, dist_config,
+                        )
+                    )
+                gossip_ps_weight.copy_(ps_weight)
+                gossip_ps_factor.copy_(ps_factor)
+            except RuntimeError as e:
+                logger.warning(f"received runtime error {e}")
+                gossiper.clean_msg_buffers_()
+                gossip_ps_weight.fill_(-1)
+            finally:
+                if gossip_stream is not None:
+                    gossip_stream.synchronize()
+                train_flag.clear()
+                gossip_flag.set()
+
+    # -- hooks ----------------------------------------------------------------
+
+    def __register_hooks(self):
+        """Forward-pre and backward hooks driving the push-sum state
+        machine (reference distributed.py:512-589).
+
+        The backward work (intra-node grad reduce + re-bias) must run at
+        the END of the whole backward pass, so the forward hook attaches a
+        grad hook to the module output which queues an autograd-engine
+        callback (the reference used the deprecated
+        ``register_backward_hook`` for the same queueing trick,
+        distributed.py:567-569)."""
+        self.register_forward_pre_hook(self.__make_forward_pre_hook())
+        queue_hook = self.__make_backward_hook()
+
+        def attach(module, inputs, output):
+            out = output
+            if isinstance(out, (tuple, list)):
+                out = next(
+                    (t for t in out if torch.is_tensor(t) and t.requires_grad),
+                    None,
+                )
+            if torch.is_tensor(out) and out.requires_grad:
+                out.register_hook(queue_hook)
+            return output
+
+        self.register_forward_hook(attach)
+
+    def __make_backward_hook(self):
+        def hook(*unused):
+            if self.nprocs_per_node > 1 and self.local_node_group is not None:
+                # intra-node grad averaging over the local group; one
+                # all-reduce when grads are flat (reference
+                # distributed.py:551-562 flattened per dtype each step)
+                if self._flatten_grads and self.flatp.flat_grad is not None:
+                    self.flatp.flat_grad.div_(self.nprocs_per_node)
+                    dist.all_reduce(
+                        self.flatp.flat_grad, group=self.local_node_group
+                    )
+                else:
+                    import functools
+
+                    grads = [
+                        p.grad.data
+                        for p in self.module.parameters()
+                        if p.requires_grad and p.grad is not None
+                    ]
+                    for g in grads:
+                        g.div_(self.nprocs_per_node)
+                    communicate(
+                        grads,
+                        functools.partial(
+                            dist.all_reduce, group=self.local_node_group
+                        ),
+                    )
+            # re-bias params before the optimizer step (reference
+            # distributed.py:565 — SGD acts on the numerator)
+            self.ps_numerator()
+
+        def queue_hook(*unused):
+            Variable._execution_engine.queue_callback(hook)
+
+        return queue_hook
+
+    def __make_forward_pre_hook(self):
+        def hook(*unused):
+            if self._flatten_grads:
+                # re-wire grads if an optimizer's zero_grad(set_to_none)
+                # detached them from the flat buffer
+                p0 = self.flatp.params[0]
+                if p0.grad is None or (
+                    p0.grad.data_ptr() != self.flatp.flat_grad.data_ptr()
+                ):
+                    self.flatp.rewire_grads()
+            if self.gossip_enable:
+                non_blocking = self.num_updates < self.synch_freq
+                if self._query_gossip_queue(non_blocking):
+                    self.num_updates = 0
+                else:
+                    self.num_updates += 1
+                if self.overlap:
+                    self.transfer_params()
+            self.unbias()
+
+        return hook

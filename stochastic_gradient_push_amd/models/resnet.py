@@ -1,0 +1,173 @@
+"""ResNet family for the ImageNet workload.
+
+The reference consumed ``torchvision.models.resnet50`` (reference
+gossip_sgd.py:693-707); this framework ships its own implementation (the
+environment has no torchvision) with identical architecture and the same
+initialization recipe the reference applied for large-batch training
+("ImageNet in 1 hour"): kaiming-normal convs, zero-init of the final
+BatchNorm gamma in every residual branch, and N(0, 0.01) for the FC layer
+(reference gossip_sgd.py:698-707).
+
+Layouts: call ``model.to(memory_format=torch.channels_last)`` on MI355X —
+MIOpen's NHWC convolutions are the fast path for bf16.
+"""
+
+from typing import List, Optional, Type, Union
+
+import torch
+import torch.nn as nn
+
+
+def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+
+
+def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, cin, planes, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = conv3x3(cin, planes, stride)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.conv2 = conv3x3(planes, planes)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, cin, planes, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = conv1x1(cin, planes)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = conv3x3(planes, planes, stride)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = conv1x1(planes, planes * self.expansion)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class ResNet(nn.Module):
+    def __init__(
+        self,
+        block: Type[Union[BasicBlock, Bottleneck]],
+        layers: List[int],
+        num_classes: int = 1000,
+        zero_init_residual: bool = True,
+    ):
+        super().__init__()
+        self.inplanes = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(block, 64, layers[0])
+        self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
+        self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
+        self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(512 * block.expansion, num_classes)
+
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(
+                    m.weight, mode="fan_out", nonlinearity="relu"
+                )
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+
+        # large-batch recipe (reference gossip_sgd.py:698-707): zero-init
+        # the last BN gamma of each residual branch; fc ~ N(0, 0.01)
+        if zero_init_residual:
+            for m in self.modules():
+                if isinstance(m, Bottleneck):
+                    nn.init.zeros_(m.bn3.weight)
+                elif isinstance(m, BasicBlock):
+                    nn.init.zeros_(m.bn2.weight)
+        nn.init.normal_(self.fc.weight, mean=0.0, std=0.01)
+        nn.init.zeros_(self.fc.bias)
+
+    def _make_layer(self, block, planes, blocks, stride=1):
+        downsample = None
+        if stride != 1 or self.inplanes != planes * block.expansion:
+            downsample = nn.Sequential(
+                conv1x1(self.inplanes, planes * block.expansion, stride),
+                nn.BatchNorm2d(planes * block.expansion),
+            )
+        layers = [block(self.inplanes, planes, stride, downsample)]
+        self.inplanes = planes * block.expansion
+        for _ in range(1, blocks):
+            layers.append(block(self.inplanes, planes))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x)
+        x = torch.flatten(x, 1)
+        return self.fc(x)
+
+
+_CONFIGS = {
+    "resnet18": (BasicBlock, [2, 2, 2, 2]),
+    "resnet34": (BasicBlock, [3, 4, 6, 3]),
+    "resnet50": (Bottleneck, [3, 4, 6, 3]),
+    "resnet101": (Bottleneck, [3, 4, 23, 3]),
+    "resnet152": (Bottleneck, [3, 8, 36, 3]),
+}
+
+
+def build_resnet(
+    name: str, num_classes: int = 1000, zero_init_residual: bool = True
+) -> ResNet:
+    block, layers = _CONFIGS[name]
+    return ResNet(block, layers, num_classes, zero_init_residual)
+
+
+def resnet18(**kw) -> ResNet:
+    return build_resnet("resnet18", **kw)
+
+
+def resnet34(**kw) -> ResNet:
+    return build_resnet("resnet34", **kw)
+
+
+def resnet50(**kw) -> ResNet:
+    return build_resnet("resnet50", **kw)
+
+
+def resnet101(**kw) -> ResNet:
+    return build_resnet("resnet101", **kw)
+
+
+def resnet152(**kw) -> ResNet:
+    return build_resnet("resnet152", **kw)

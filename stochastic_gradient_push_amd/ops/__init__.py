@@ -1,0 +1,154 @@
+"""Fused flat-buffer device ops.
+
+The gossip hot path of the reference is ~161 tiny per-tensor elementwise
+CUDA ops per step (reference gossip/distributed.py:298-314, 372-379,
+402-425; gossip/ad_psgd.py:357-361).  Here every hot op is a single fused
+HIP/CDNA4 kernel launch over one contiguous flat buffer
+(`csrc/gossip_kernels.hip`, built for gfx950):
+
+==================  =====================================================
+op                  semantics (all in-place over flat 1-D buffers)
+==================  =====================================================
+scale_              x *= a                  (bias / de-bias,
+                                             reference distributed.py:302-313)
+add_scale_          x = (x + r) * a         (residual accumulate + lazy
+                                             mix, reference distributed.py:372-379)
+pack_mix_           x *= a; out = x         (transfer: scale params and
+                                             pack comm buffer in one pass,
+                                             reference distributed.py:409-418)
+average_            x = (x + y) * 0.5       (bilateral avg,
+                                             reference ad_psgd.py:357-361)
+sgd_step_           fused momentum-SGD update (reference relies on
+                                             torch.optim.SGD, ad_psgd.py:261-266)
+==================  =====================================================
+
+Scalars ``a`` may be Python floats or 1-element device tensors (no host
+sync on the training path: push-sum weights stay device-resident).
+
+Dispatch: CUDA(HIP) tensors require the compiled ``_gossip_kernels``
+extension — a missing extension on a GPU box raises immediately rather
+than silently falling back to eager.  CPU tensors use the PyTorch
+reference implementations (these are also the numerics oracle in tests).
+"""
+
+from typing import Optional, Union
+
+import torch
+
+Scalar = Union[float, torch.Tensor]
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import _gossip_kernels  # in-tree built .so
+
+        _EXT = _gossip_kernels
+    except ImportError as e:  # pragma: no cover - exercised on GPU boxes
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def extension_available() -> bool:
+    return _load_extension() is not None
+
+
+def _ext_for(t: torch.Tensor):
+    """Return the extension for a CUDA tensor, or raise loudly."""
+    ext = _load_extension()
+    if ext is None:
+        raise RuntimeError(
+            "stochastic_gradient_push_amd HIP extension (_gossip_kernels) is "
+            "not built but a CUDA tensor reached the fused-op path. Build it "
+            f"with `python setup.py build_ext --inplace`. Import error: {_EXT_ERR}"
+        )
+    return ext
+
+
+def _as_scalar_tensor(a: Scalar, like: torch.Tensor) -> torch.Tensor:
+    if isinstance(a, torch.Tensor):
+        return a
+    return torch.tensor([float(a)], device=like.device, dtype=torch.float32)
+
+
+def scale_(x: torch.Tensor, a: Scalar) -> torch.Tensor:
+    """x *= a (fused bias/de-bias over the whole flat parameter buffer)."""
+    if x.is_cuda:
+        _ext_for(x).scale_(x, _as_scalar_tensor(a, x))
+    else:
+        x.mul_(a if isinstance(a, float) else a.to(x.dtype))
+    return x
+
+
+def add_scale_(x: torch.Tensor, r: torch.Tensor, a: Scalar = 1.0) -> torch.Tensor:
+    """x = (x + r) * a in a single pass (residual accumulate + lazy mix)."""
+    if x.is_cuda:
+        _ext_for(x).add_scale_(x, r, _as_scalar_tensor(a, x))
+    else:
+        x.add_(r)
+        if not (isinstance(a, float) and a == 1.0):
+            x.mul_(a if isinstance(a, float) else a.to(x.dtype))
+    return x
+
+
+def pack_mix_(x: torch.Tensor, out: torch.Tensor, a: Scalar = 1.0) -> torch.Tensor:
+    """x *= a; out = x — one read of x, two writes (transfer_params path)."""
+    if x.is_cuda:
+        _ext_for(x).pack_mix_(x, out, _as_scalar_tensor(a, x))
+    else:
+        if not (isinstance(a, float) and a == 1.0):
+            x.mul_(a if isinstance(a, float) else a.to(x.dtype))
+        out.copy_(x)
+    return out
+
+
+def average_(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """x = (x + y) * 0.5 (bilateral gossip merge)."""
+    if x.is_cuda:
+        _ext_for(x).average_(x, y)
+    else:
+        x.add_(y).mul_(0.5)
+    return x
+
+
+def sgd_step_(
+    params: torch.Tensor,
+    grads: torch.Tensor,
+    momentum_buf: torch.Tensor,
+    lr: float,
+    momentum: float = 0.0,
+    weight_decay: float = 0.0,
+    dampening: float = 0.0,
+    nesterov: bool = False,
+    first_step: bool = False,
+) -> None:
+    """Fused SGD-with-momentum update over flat param/grad buffers.
+
+    Semantics match torch.optim.SGD:
+      d = g + wd * p
+      buf = momentum * buf + (1 - dampening) * d   (buf = d on first step)
+      d = d + momentum * buf  (nesterov)  |  d = buf  (plain momentum)
+      p -= lr * d
+    """
+    if params.is_cuda:
+        _ext_for(params).sgd_step_(
+            params, grads, momentum_buf, float(lr), float(momentum),
+            float(weight_decay), float(dampening), bool(nesterov),
+            bool(first_step),
+        )
+        return
+    d = grads
+    if weight_decay != 0.0:
+        d = d.add(params, alpha=weight_decay)
+    if momentum != 0.0:
+        if first_step:
+            momentum_buf.copy_(d)
+        else:
+            momentum_buf.mul_(momentum).add_(d, alpha=1.0 - dampening)
+        d = d.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
+    params.add_(d, alpha=-lr)

@@ -1,0 +1,74 @@
+"""flatten/unflatten round-trips, Meter stats, is_power_of."""
+
+import math
+
+import pytest
+import torch
+
+from stochastic_gradient_push_amd.utils import (
+    Meter,
+    flatten_tensors,
+    group_by_dtype,
+    is_power_of,
+    unflatten_tensors,
+)
+
+
+def test_flatten_unflatten_roundtrip():
+    tensors = [torch.randn(3, 4), torch.randn(7), torch.randn(2, 2, 2)]
+    flat = flatten_tensors(tensors)
+    assert flat.numel() == sum(t.numel() for t in tensors)
+    out = unflatten_tensors(flat, tensors)
+    for a, b in zip(tensors, out):
+        assert torch.equal(a, b)
+
+
+def test_flatten_single():
+    t = [torch.randn(5, 5)]
+    flat = flatten_tensors(t)
+    assert flat.shape == (25,)
+    # must be a copy, not a view
+    flat.add_(1)
+    assert not torch.allclose(flat.view(5, 5), t[0])
+
+
+def test_group_by_dtype():
+    ts = [torch.randn(2), torch.randn(2).double(), torch.randn(3)]
+    g = group_by_dtype(ts)
+    assert len(g[torch.float32]) == 2
+    assert len(g[torch.float64]) == 1
+
+
+def test_meter_stats():
+    m = Meter(stateful=True)
+    vals = [1.0, 2.0, 3.0, 4.0]
+    for v in vals:
+        m.update(v)
+    assert m.avg == pytest.approx(2.5)
+    mean = sum(vals) / 4
+    std = (sum((v - mean) ** 2 for v in vals) / 3) ** 0.5
+    assert m.std == pytest.approx(std)
+    assert m.mad == pytest.approx(1.0)
+    assert str(m) == "4.000,2.500,1.000"
+
+
+def test_meter_csv_and_pretty():
+    m = Meter(ptag="BT", stateful=False, csv_format=False)
+    m.update(1.5)
+    assert str(m).startswith("BT: 1.500")
+
+
+def test_meter_state_dict_roundtrip():
+    m = Meter(stateful=True)
+    for v in (1.0, 5.0):
+        m.update(v)
+    m2 = Meter(init_dict=m.state_dict(), stateful=True)
+    assert m2.avg == m.avg and m2.count == m.count
+
+
+def test_is_power_of():
+    assert is_power_of(8, 2)
+    assert is_power_of(27, 3)
+    assert not is_power_of(12, 2)
+    assert is_power_of(1, 0)
+    assert not is_power_of(5, 1)
