@@ -1,0 +1,201 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 gossip-SGP training throughput on MI355X.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+For N > 1 the driver launches via torch.distributed.run (one rank per
+GPU over RCCL).  Rank 0 prints exactly one JSON line with the whole-job
+images/sec (BASELINE.json metric: images/sec whole node, ResNet-50 SGP,
+synthetic 224x224, random init) plus the per-step gossip milliseconds.
+
+Weak scaling: per-GPU batch fixed (default 32 — the reference recipe's
+256/node over 8 GPUs, job_scripts/*.sh), so global batch grows with N.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=32, help="per-GPU batch")
+    p.add_argument("--model", type=str, default="resnet50")
+    p.add_argument(
+        "--algorithm", type=str, default="sgp",
+        choices=["sgp", "osgp", "dpsgd", "ar", "adpsgd"],
+    )
+    p.add_argument("--peers-per-itr", type=int, default=1)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--device", type=str, default="cuda")
+    p.add_argument("--no-channels-last", action="store_true")
+    return p.parse_args()
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, flush=True)
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    cuda = args.device == "cuda"
+
+    if cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if world_size > 1:
+        backend = "nccl" if cuda else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from stochastic_gradient_push_amd import (
+        GossipDataParallel,
+        NPeerDynamicDirectedExponentialGraph,
+    )
+    from stochastic_gradient_push_amd.models import build_resnet
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(1234 + rank)
+    model = build_resnet(args.model).to(device)
+    if cuda and not args.no_channels_last:
+        model = model.to(memory_format=torch.channels_last)
+
+    use_ddp = args.algorithm == "ar" and world_size > 1
+    if use_ddp:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank] if cuda else None,
+            bucket_cap_mb=64,
+        )
+        opt = torch.optim.SGD(
+            model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4,
+        )
+        gdp = None
+    else:
+        graph = (
+            NPeerDynamicDirectedExponentialGraph(
+                rank, world_size, peers_per_itr=args.peers_per_itr
+            )
+            if world_size > 1 else None
+        )
+        gdp = GossipDataParallel(
+            model,
+            graph=graph,
+            push_sum=args.algorithm in ("sgp", "osgp"),
+            overlap=args.algorithm == "osgp",
+            rank=rank if world_size > 1 else 0,
+            world_size=world_size,
+        )
+        model = gdp
+        opt = FusedSGD(
+            gdp.flatp, lr=0.1, momentum=0.9, weight_decay=1e-4,
+        )
+
+    loss_fn = nn.CrossEntropyLoss()
+    model.train()
+
+    x = torch.randn(args.batch_size, 3, 224, 224, device=device)
+    if cuda and not args.no_channels_last:
+        x = x.to(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (args.batch_size,), device=device)
+
+    amp_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    use_amp = args.dtype == "bf16"
+
+    def step():
+        with torch.autocast(
+            device_type="cuda" if cuda else "cpu",
+            dtype=amp_dtype, enabled=use_amp,
+        ):
+            out = model(x)
+            loss = loss_fn(out, y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        if gdp is not None and args.algorithm != "osgp":
+            gdp.transfer_params()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    def barrier_sync():
+        if world_size > 1:
+            dist.barrier()
+        if cuda:
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max elapsed over ranks
+    if world_size > 1:
+        t = torch.tensor([elapsed], device=device if cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world_size if world_size > 1 else (1 if cuda else 0) or 1
+    global_batch = args.batch_size * world_size
+    images_per_sec = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    gossip_ms = gdp.gossip_ms() if gdp is not None else 0.0
+    if world_size > 1 and gdp is not None:
+        g = torch.tensor([gossip_ms], device=device if cuda else "cpu")
+        dist.all_reduce(g, op=dist.ReduceOp.MAX)
+        gossip_ms = g.item()
+
+    if rank == 0:
+        result = {
+            "metric": "images/sec",
+            "value": round(images_per_sec, 2),
+            "unit": "images/sec",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": None,
+                "image_size": 224,
+                "parallelism": f"{args.algorithm}-dp{world_size}",
+                "peers_per_itr": args.peers_per_itr,
+                "gossip_ms_per_step": round(gossip_ms, 3),
+                "channels_last": not args.no_channels_last,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if gdp is not None:
+        gdp.shutdown()
+    if world_size > 1:
+        dist.barrier()
+        dist.destroy_process_group()
+    os._exit(0)
+
+
+if __name__ == "__main__":
+    main()

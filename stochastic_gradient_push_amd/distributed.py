@@ -113,7 +113,10 @@ class GossipDataParallel(Module):
 
         # communication device (reference distributed.py:101-105)
         if comm_device is None:
-            cpu_comm = dist.get_backend() == "gloo"
+            if dist.is_initialized():
+                cpu_comm = dist.get_backend() == "gloo"
+            else:
+                cpu_comm = not self._cuda
             comm_device = torch.device("cpu") if cpu_comm else torch.device("cuda")
         self.__cpu_comm = comm_device.type == "cpu"
 
@@ -189,6 +192,7 @@ class GossipDataParallel(Module):
         self.gossip_lock = threading.Lock()
         self.gossip_flag = threading.Event()
         self.train_flag = threading.Event()
+        self.stop_flag = threading.Event()
         if self._cuda and not self.__cpu_comm and use_streams:
             self.gossip_stream = torch.cuda.Stream()
         elif self._cuda:
@@ -210,6 +214,7 @@ class GossipDataParallel(Module):
                     self.gossip_ps_factor,
                     self.gossip_stream,
                     self.gossip_group,
+                    self.stop_flag,
                 ),
                 daemon=True,
                 name="Gossip-Thread",
@@ -293,6 +298,23 @@ class GossipDataParallel(Module):
 
     def sync_comms(self):
         self._query_gossip_queue(non_blocking=False)
+
+    def shutdown(self, timeout: float = 10.0):
+        """Stop the background gossip thread cleanly (drains any in-flight
+        exchange first).  Call before ``dist.destroy_process_group()``."""
+        if not self.distributed or self.gossip_thread is None:
+            return
+        if self.gossiping:
+            self.gossip_flag.wait(timeout=timeout)
+        self.stop_flag.set()
+        self.train_flag.set()  # wake the thread so it can observe stop
+        self.gossip_thread.join(timeout=timeout)
+
+    def gossip_ms(self) -> float:
+        """Average wall-clock milliseconds of one gossip exchange (the
+        north-star 'per-step gossip ms' metric)."""
+        meter = self.dist_config.get("gossip_meter")
+        return meter.avg * 1000.0 if meter is not None and meter.count else 0.0
 
     # -- intra-node multiprocess tier ---------------------------------------
 
@@ -450,10 +472,16 @@ class GossipDataParallel(Module):
     def _gossip_target(
         dist_config, gossip_flag, train_flag, gossip_lock, gossip_params,
         gossip_device_buffer, gossip_ps_weight, gossip_ps_factor,
-        gossip_stream, gossip_group,
+        gossip_stream, gossip_group, stop_flag,
     ):
         """Background gossip loop (reference distributed.py:459-510)."""
+        import time as _time
+
+        from .utils.metering import Meter
+
         logger = make_logger(dist_config["rank"], dist_config["verbose"])
+        gossip_meter = Meter(ptag="Gossip", stateful=False, csv_format=False)
+        dist_config["gossip_meter"] = gossip_meter
 
         gossiper_class = PushSum if dist_config["push_sum"] else PushPull
         gossiper = gossiper_class(
@@ -472,7 +500,11 @@ class GossipDataParallel(Module):
 
         while True:
             train_flag.wait()
+            if stop_flag.is_set():
+                logger.debug("gossip thread stopping")
+                return
             logger.debug("received train-flag")
+            _t0 = _time.perf_counter()
             try:
                 if gossip_stream is not None:
                     with torch.cuda.stream(gossip_stream):
@@ -498,6 +530,7 @@ class GossipDataParallel(Module):
             finally:
                 if gossip_stream is not None:
                     gossip_stream.synchronize()
+                gossip_meter.update(_time.perf_counter() - _t0)
                 train_flag.clear()
                 gossip_flag.set()
 
