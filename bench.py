@@ -37,6 +37,10 @@ def parse_args():
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default="cuda")
     p.add_argument("--no-channels-last", action="store_true")
+    p.add_argument(
+        "--no-graph", action="store_true",
+        help="disable hipGraph capture of the local train step",
+    )
     return p.parse_args()
 
 
@@ -55,6 +59,8 @@ def main():
     if cuda:
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
+        # let MIOpen search for the fastest conv kernels (fixed shapes)
+        torch.backends.cudnn.benchmark = True
     else:
         device = torch.device("cpu")
 
@@ -115,19 +121,48 @@ def main():
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     use_amp = args.dtype == "bf16"
 
-    def step():
+    def compute_step(net):
+        """Local fwd+bwd+optimizer (everything but gossip)."""
         with torch.autocast(
             device_type="cuda" if cuda else "cpu",
             dtype=amp_dtype, enabled=use_amp,
         ):
-            out = model(x)
+            out = net(x)
             loss = loss_fn(out, y)
         loss.backward()
         opt.step()
         opt.zero_grad()
-        if gdp is not None and args.algorithm != "osgp":
-            gdp.transfer_params()
         return loss
+
+    # hipGraph capture of the launch-bound local step.  Requires lazy
+    # mixing (bias/de-bias are no-ops) so the graphed region is pure
+    # compute; gossip (p2p comm + fused residual merge + host flag logic)
+    # runs around the replay.
+    use_graph = (
+        cuda and not args.no_graph and not use_ddp
+        and (gdp is None or gdp.lazy_mixing)
+    )
+    if use_graph:
+        inner = gdp.module if gdp is not None else model
+        for _ in range(3):
+            compute_step(inner)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            compute_step(inner)
+
+        def step():
+            graph.replay()
+            if gdp is not None:
+                gdp._query_gossip_queue(non_blocking=gdp.asynch)
+                gdp.transfer_params()
+            return None
+    else:
+        def step():
+            loss = compute_step(model)
+            if gdp is not None and args.algorithm != "osgp":
+                gdp.transfer_params()
+            return loss
 
     for _ in range(args.warmup):
         step()
