@@ -229,7 +229,6 @@ def main():
     if world_size > 1:
         dist.barrier()
         dist.destroy_process_group()
-    os._exit(0)
 
 
 if __name__ == "__main__":
