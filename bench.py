@@ -34,6 +34,10 @@ def parse_args():
         choices=["sgp", "osgp", "dpsgd", "ar", "adpsgd"],
     )
     p.add_argument("--peers-per-itr", type=int, default=1)
+    p.add_argument(
+        "--norm", type=str, default="miopen",
+        help="batch-norm backend: miopen|native|fused",
+    )
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default="cuda")
     p.add_argument("--no-channels-last", action="store_true")
@@ -76,7 +80,7 @@ def main():
     from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
 
     torch.manual_seed(1234 + rank)
-    model = build_resnet(args.model).to(device)
+    model = build_resnet(args.model, norm=args.norm).to(device)
     if cuda and not args.no_channels_last:
         model = model.to(memory_format=torch.channels_last)
 
@@ -220,6 +224,7 @@ def main():
                 "peers_per_itr": args.peers_per_itr,
                 "gossip_ms_per_step": round(gossip_ms, 3),
                 "channels_last": not args.no_channels_last,
+                "norm": args.norm,
             },
         }
         print(json.dumps(result), flush=True)

@@ -29,13 +29,14 @@ def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
 class BasicBlock(nn.Module):
     expansion = 1
 
-    def __init__(self, cin, planes, stride=1, downsample=None):
+    def __init__(self, cin, planes, stride=1, downsample=None,
+                 norm_layer=nn.BatchNorm2d):
         super().__init__()
         self.conv1 = conv3x3(cin, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = norm_layer(planes)
         self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = norm_layer(planes)
         self.downsample = downsample
         self.stride = stride
 
@@ -52,14 +53,15 @@ class BasicBlock(nn.Module):
 class Bottleneck(nn.Module):
     expansion = 4
 
-    def __init__(self, cin, planes, stride=1, downsample=None):
+    def __init__(self, cin, planes, stride=1, downsample=None,
+                 norm_layer=nn.BatchNorm2d):
         super().__init__()
         self.conv1 = conv1x1(cin, planes)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = norm_layer(planes)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = norm_layer(planes)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.bn3 = norm_layer(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
         self.stride = stride
@@ -82,11 +84,15 @@ class ResNet(nn.Module):
         layers: List[int],
         num_classes: int = 1000,
         zero_init_residual: bool = True,
+        norm: str = "miopen",
     ):
         super().__init__()
+        from .layers import make_norm
+
+        self._norm_layer = make_norm(norm)
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = self._norm_layer(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, layers[0])
@@ -121,12 +127,14 @@ class ResNet(nn.Module):
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = nn.Sequential(
                 conv1x1(self.inplanes, planes * block.expansion, stride),
-                nn.BatchNorm2d(planes * block.expansion),
+                self._norm_layer(planes * block.expansion),
             )
-        layers = [block(self.inplanes, planes, stride, downsample)]
+        layers = [block(self.inplanes, planes, stride, downsample,
+                        norm_layer=self._norm_layer)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
-            layers.append(block(self.inplanes, planes))
+            layers.append(block(self.inplanes, planes,
+                                norm_layer=self._norm_layer))
         return nn.Sequential(*layers)
 
     def forward(self, x):
@@ -147,10 +155,11 @@ _CONFIGS = {
 
 
 def build_resnet(
-    name: str, num_classes: int = 1000, zero_init_residual: bool = True
+    name: str, num_classes: int = 1000, zero_init_residual: bool = True,
+    norm: str = "miopen",
 ) -> ResNet:
     block, layers = _CONFIGS[name]
-    return ResNet(block, layers, num_classes, zero_init_residual)
+    return ResNet(block, layers, num_classes, zero_init_residual, norm=norm)
 
 
 def resnet18(**kw) -> ResNet:
