@@ -24,6 +24,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "gossip_kernels.hip"),
+        os.path.join(CSRC, "bn_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -47,3 +47,120 @@ def make_norm(kind: str):
     if kind == "native":
         return NativeBatchNorm2d
     raise ValueError(f"unknown norm kind {kind}")
+
+
+def _fused_supported(x: torch.Tensor, C: int) -> bool:
+    """The hand-written kernels need bf16 NHWC with C = 8 * 2^k <= 2048
+    (the reduce kernel's thread geometry; every ResNet width qualifies)."""
+    cdiv8 = C // 8
+    return (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and C % 8 == 0
+        and cdiv8 <= 256
+        and (cdiv8 & (cdiv8 - 1)) == 0
+        and x.is_contiguous(memory_format=torch.channels_last)
+    )
+
+
+class _FusedBNFunction(torch.autograd.Function):
+    """Fused NHWC BN(+residual)(+ReLU), bf16 activations / fp32 stats.
+
+    Forward: 1 reduce + 1 finalize + 1 apply kernel (training).
+    Backward: 1 reduce (dgamma/dbeta) + 1 apply (dx [, dresidual]).
+    """
+
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
+                momentum, eps, training, relu):
+        from .. import ops as _o
+
+        k = _o._ext_for(x)
+        N, C, H, W = x.shape
+        M = N * H * W
+        y = torch.empty_like(x)
+        scale_shift = torch.empty(2 * C, device=x.device, dtype=torch.float32)
+        if training:
+            scratch = torch.zeros(2 * C, device=x.device, dtype=torch.float32)
+            smean = torch.empty(C, device=x.device, dtype=torch.float32)
+            sinvstd = torch.empty(C, device=x.device, dtype=torch.float32)
+            k.bn_fwd_reduce(x, scratch, M, C)
+            k.bn_fwd_finalize(
+                scratch, gamma, beta, running_mean, running_var, smean,
+                sinvstd, scale_shift, momentum, eps, M, C, True,
+            )
+        else:
+            k.bn_eval_prep(running_mean, running_var, gamma, beta,
+                           scale_shift, eps, C)
+            smean = running_mean
+            sinvstd = torch.rsqrt(running_var + eps)
+        k.bn_fwd_apply(x, residual, y, scale_shift, M, C, relu)
+        ctx.save_for_backward(x, y, gamma, smean, sinvstd)
+        ctx.bn_shape = (M, C)
+        ctx.bn_relu = relu
+        ctx.bn_training = training
+        ctx.bn_has_res = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import ops as _o
+
+        x, y, gamma, smean, sinvstd = ctx.saved_tensors
+        k = _o._ext_for(x)
+        M, C = ctx.bn_shape
+        relu = ctx.bn_relu
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        scratch = torch.zeros(2 * C, device=x.device, dtype=torch.float32)
+        k.bn_bwd_reduce(x, dy, y if relu else None, smean, sinvstd,
+                        scratch, M, C, relu)
+        dx = torch.empty_like(x)
+        dres = torch.empty_like(x) if ctx.bn_has_res else None
+        if ctx.bn_training:
+            k.bn_bwd_apply(x, dy, y if relu else None, dx, dres, smean,
+                           sinvstd, gamma, scratch, M, C, relu)
+        else:
+            k.bn_bwd_apply_eval(dy, y if relu else None, dx, dres, sinvstd,
+                                gamma, M, C, relu)
+        return (dx, dres, scratch[:C], scratch[C:], None, None, None, None,
+                None, None)
+
+
+class FusedBatchNorm2d(nn.BatchNorm2d):
+    """BatchNorm2d[+residual add][+ReLU] in hand-written CDNA4 kernels.
+
+    ``forward(x, residual=None)``: when ``residual`` is given it is added
+    after normalization, before the (optional) ReLU — the ResNet residual
+    join.  Falls back to the equivalent PyTorch ops for inputs the
+    kernels don't cover (CPU, fp32, non-channels-last).
+    """
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False):
+        super().__init__(
+            num_features, eps=eps, momentum=momentum, affine=True,
+            track_running_stats=True,
+        )
+        self.relu = relu
+        self.num_batches_tracked = None
+
+    def forward(self, x, residual=None):
+        if _fused_supported(x, self.num_features) and (
+            residual is None
+            or residual.is_contiguous(memory_format=torch.channels_last)
+        ):
+            with torch.amp.autocast(device_type="cuda", enabled=False):
+                return _FusedBNFunction.apply(
+                    x, residual, self.weight, self.bias, self.running_mean,
+                    self.running_var, self.momentum, self.eps,
+                    self.training, self.relu,
+                )
+        # reference fallback path (also the CPU numerics oracle)
+        y = torch.batch_norm(
+            x, self.weight, self.bias, self.running_mean, self.running_var,
+            self.training, self.momentum, self.eps, False,
+        )
+        if residual is not None:
+            y = y + residual
+        if self.relu:
+            y = torch.relu(y)
+        return y

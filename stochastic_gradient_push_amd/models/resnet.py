@@ -30,17 +30,29 @@ class BasicBlock(nn.Module):
     expansion = 1
 
     def __init__(self, cin, planes, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d):
+                 norm_layer=nn.BatchNorm2d, fused=False):
         super().__init__()
+        self._fused = fused
         self.conv1 = conv3x3(cin, planes, stride)
-        self.bn1 = norm_layer(planes)
-        self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = norm_layer(planes)
         self.downsample = downsample
         self.stride = stride
+        if fused:
+            from .layers import FusedBatchNorm2d
+
+            self.bn1 = FusedBatchNorm2d(planes, relu=True)
+            # bn2 fuses the residual add + final relu
+            self.bn2 = FusedBatchNorm2d(planes, relu=True)
+        else:
+            self.bn1 = norm_layer(planes)
+            self.bn2 = norm_layer(planes)
+            self.relu = nn.ReLU(inplace=True)
 
     def forward(self, x):
+        if self._fused:
+            out = self.bn1(self.conv1(x))
+            identity = x if self.downsample is None else self.downsample(x)
+            return self.bn2(self.conv2(out), identity)
         identity = x
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.bn2(self.conv2(out))
@@ -54,19 +66,33 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, cin, planes, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d):
+                 norm_layer=nn.BatchNorm2d, fused=False):
         super().__init__()
+        self._fused = fused
         self.conv1 = conv1x1(cin, planes)
-        self.bn1 = norm_layer(planes)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = norm_layer(planes)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = norm_layer(planes * self.expansion)
-        self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
         self.stride = stride
+        if fused:
+            from .layers import FusedBatchNorm2d
+
+            self.bn1 = FusedBatchNorm2d(planes, relu=True)
+            self.bn2 = FusedBatchNorm2d(planes, relu=True)
+            # bn3 fuses the residual add + final relu
+            self.bn3 = FusedBatchNorm2d(planes * self.expansion, relu=True)
+        else:
+            self.bn1 = norm_layer(planes)
+            self.bn2 = norm_layer(planes)
+            self.bn3 = norm_layer(planes * self.expansion)
+            self.relu = nn.ReLU(inplace=True)
 
     def forward(self, x):
+        if self._fused:
+            out = self.bn1(self.conv1(x))
+            out = self.bn2(self.conv2(out))
+            identity = x if self.downsample is None else self.downsample(x)
+            return self.bn3(self.conv3(out), identity)
         identity = x
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
@@ -87,12 +113,19 @@ class ResNet(nn.Module):
         norm: str = "miopen",
     ):
         super().__init__()
-        from .layers import make_norm
+        from .layers import FusedBatchNorm2d, make_norm
 
-        self._norm_layer = make_norm(norm)
+        self._fused = norm == "fused"
+        self._norm_layer = (
+            (lambda c: FusedBatchNorm2d(c, relu=False))
+            if self._fused else make_norm(norm)
+        )
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = self._norm_layer(64)
+        self.bn1 = (
+            FusedBatchNorm2d(64, relu=True) if self._fused
+            else self._norm_layer(64)
+        )
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, layers[0])
@@ -130,15 +163,19 @@ class ResNet(nn.Module):
                 self._norm_layer(planes * block.expansion),
             )
         layers = [block(self.inplanes, planes, stride, downsample,
-                        norm_layer=self._norm_layer)]
+                        norm_layer=self._norm_layer, fused=self._fused)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
             layers.append(block(self.inplanes, planes,
-                                norm_layer=self._norm_layer))
+                                norm_layer=self._norm_layer,
+                                fused=self._fused))
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        if self._fused:
+            x = self.maxpool(self.bn1(self.conv1(x)))  # relu fused in bn1
+        else:
+            x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x)
         x = torch.flatten(x, 1)

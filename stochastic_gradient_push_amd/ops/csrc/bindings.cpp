@@ -6,7 +6,34 @@
 
 #include <c10/hip/HIPStream.h>
 
+typedef unsigned short ushort_t;
+
 extern "C" {
+void bn_fwd_reduce(const ushort_t* x, float* scratch, int64_t M, int C,
+                   hipStream_t s);
+void bn_fwd_finalize(const float* scratch, const float* gamma,
+                     const float* beta, float* rmean, float* rvar,
+                     float* smean, float* sinvstd, float* scale_shift,
+                     double momentum, double eps, int64_t M, int C,
+                     bool update_running, hipStream_t s);
+void bn_eval_prep(const float* rmean, const float* rvar, const float* gamma,
+                  const float* beta, float* scale_shift, double eps, int C,
+                  hipStream_t s);
+void bn_fwd_apply(const ushort_t* x, const ushort_t* res, ushort_t* y,
+                  const float* scale_shift, int64_t M, int C, bool relu,
+                  hipStream_t s);
+void bn_bwd_reduce(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
+                   const float* smean, const float* sinvstd, float* scratch,
+                   int64_t M, int C, bool relu, hipStream_t s);
+void bn_bwd_apply(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
+                  ushort_t* dx, ushort_t* dres, const float* smean,
+                  const float* sinvstd, const float* gamma,
+                  const float* scratch, int64_t M, int C, bool relu,
+                  hipStream_t s);
+void bn_bwd_apply_eval(const ushort_t* dy, const ushort_t* y, ushort_t* dx,
+                       ushort_t* dres, const float* sinvstd,
+                       const float* gamma, int64_t M, int C, bool relu,
+                       hipStream_t s);
 void sgp_scale(float* x, const float* a, int64_t n, hipStream_t stream);
 void sgp_add_scale(float* x, const float* r, const float* a, int64_t n,
                    hipStream_t stream);
@@ -83,13 +110,155 @@ void sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                p.numel(), current_stream(p));
 }
 
+// ---------------------------------------------------------------- BN ops
+
+const ushort_t* bf16_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const ushort_t*>(t.data_ptr());
+}
+ushort_t* bf16_mut(torch::Tensor& t) {
+  return reinterpret_cast<ushort_t*>(t.data_ptr());
+}
+
+void check_act(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on device");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+void check_f32(const torch::Tensor& t, const char* name, int64_t numel) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous(), name,
+              " must be contiguous device tensor");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.numel() == numel, name, " wrong size");
+}
+
+void bn_fwd_reduce_py(torch::Tensor x, torch::Tensor scratch, int64_t M,
+                      int64_t C) {
+  check_act(x, "x");
+  check_f32(scratch, "scratch", 2 * C);
+  bn_fwd_reduce(bf16_ptr(x), scratch.data_ptr<float>(), M, (int)C,
+                current_stream(x));
+}
+
+void bn_fwd_finalize_py(torch::Tensor scratch, torch::Tensor gamma,
+                        torch::Tensor beta, torch::Tensor rmean,
+                        torch::Tensor rvar, torch::Tensor smean,
+                        torch::Tensor sinvstd, torch::Tensor scale_shift,
+                        double momentum, double eps, int64_t M, int64_t C,
+                        bool update_running) {
+  check_f32(scratch, "scratch", 2 * C);
+  check_f32(gamma, "gamma", C);
+  check_f32(beta, "beta", C);
+  check_f32(rmean, "rmean", C);
+  check_f32(rvar, "rvar", C);
+  check_f32(smean, "smean", C);
+  check_f32(sinvstd, "sinvstd", C);
+  check_f32(scale_shift, "scale_shift", 2 * C);
+  bn_fwd_finalize(scratch.data_ptr<float>(), gamma.data_ptr<float>(),
+                  beta.data_ptr<float>(), rmean.data_ptr<float>(),
+                  rvar.data_ptr<float>(), smean.data_ptr<float>(),
+                  sinvstd.data_ptr<float>(), scale_shift.data_ptr<float>(),
+                  momentum, eps, M, (int)C, update_running,
+                  current_stream(scratch));
+}
+
+void bn_eval_prep_py(torch::Tensor rmean, torch::Tensor rvar,
+                     torch::Tensor gamma, torch::Tensor beta,
+                     torch::Tensor scale_shift, double eps, int64_t C) {
+  check_f32(scale_shift, "scale_shift", 2 * C);
+  bn_eval_prep(rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+               gamma.data_ptr<float>(), beta.data_ptr<float>(),
+               scale_shift.data_ptr<float>(), eps, (int)C,
+               current_stream(scale_shift));
+}
+
+void bn_fwd_apply_py(torch::Tensor x, torch::optional<torch::Tensor> res,
+                     torch::Tensor y, torch::Tensor scale_shift, int64_t M,
+                     int64_t C, bool relu) {
+  check_act(x, "x");
+  check_act(y, "y");
+  check_f32(scale_shift, "scale_shift", 2 * C);
+  const ushort_t* rp = nullptr;
+  if (res.has_value()) {
+    check_act(res.value(), "res");
+    rp = bf16_ptr(res.value());
+  }
+  bn_fwd_apply(bf16_ptr(x), rp, bf16_mut(y), scale_shift.data_ptr<float>(),
+               M, (int)C, relu, current_stream(x));
+}
+
+void bn_bwd_reduce_py(torch::Tensor x, torch::Tensor dy,
+                      torch::optional<torch::Tensor> y, torch::Tensor smean,
+                      torch::Tensor sinvstd, torch::Tensor scratch,
+                      int64_t M, int64_t C, bool relu) {
+  check_act(x, "x");
+  check_act(dy, "dy");
+  check_f32(scratch, "scratch", 2 * C);
+  const ushort_t* yp = nullptr;
+  if (relu) {
+    TORCH_CHECK(y.has_value(), "y required for relu backward");
+    check_act(y.value(), "y");
+    yp = bf16_ptr(y.value());
+  }
+  bn_bwd_reduce(bf16_ptr(x), bf16_ptr(dy), yp, smean.data_ptr<float>(),
+                sinvstd.data_ptr<float>(), scratch.data_ptr<float>(), M,
+                (int)C, relu, current_stream(x));
+}
+
+void bn_bwd_apply_py(torch::Tensor x, torch::Tensor dy,
+                     torch::optional<torch::Tensor> y, torch::Tensor dx,
+                     torch::optional<torch::Tensor> dres,
+                     torch::Tensor smean, torch::Tensor sinvstd,
+                     torch::Tensor gamma, torch::Tensor scratch, int64_t M,
+                     int64_t C, bool relu) {
+  check_act(x, "x");
+  check_act(dy, "dy");
+  check_act(dx, "dx");
+  const ushort_t* yp = nullptr;
+  ushort_t* drp = nullptr;
+  if (relu) {
+    TORCH_CHECK(y.has_value(), "y required for relu backward");
+    yp = bf16_ptr(y.value());
+  }
+  if (dres.has_value()) drp = bf16_mut(dres.value());
+  bn_bwd_apply(bf16_ptr(x), bf16_ptr(dy), yp, bf16_mut(dx), drp,
+               smean.data_ptr<float>(), sinvstd.data_ptr<float>(),
+               gamma.data_ptr<float>(), scratch.data_ptr<float>(), M,
+               (int)C, relu, current_stream(x));
+}
+
+void bn_bwd_apply_eval_py(torch::Tensor dy, torch::optional<torch::Tensor> y,
+                          torch::Tensor dx,
+                          torch::optional<torch::Tensor> dres,
+                          torch::Tensor sinvstd, torch::Tensor gamma,
+                          int64_t M, int64_t C, bool relu) {
+  check_act(dy, "dy");
+  check_act(dx, "dx");
+  const ushort_t* yp = nullptr;
+  ushort_t* drp = nullptr;
+  if (relu) {
+    TORCH_CHECK(y.has_value(), "y required for relu backward");
+    yp = bf16_ptr(y.value());
+  }
+  if (dres.has_value()) drp = bf16_mut(dres.value());
+  bn_bwd_apply_eval(bf16_ptr(dy), yp, bf16_mut(dx), drp,
+                    sinvstd.data_ptr<float>(), gamma.data_ptr<float>(), M,
+                    (int)C, relu, current_stream(dy));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.doc() = "gfx950 fused gossip kernels";
+  m.doc() = "gfx950 fused gossip + batchnorm kernels";
   m.def("scale_", &scale_, "x *= a (in place, fused over flat buffer)");
   m.def("add_scale_", &add_scale_, "x = (x + r) * a");
   m.def("pack_mix_", &pack_mix_, "x *= a; out = x");
   m.def("average_", &average_, "x = (x + y) / 2");
   m.def("sgd_step_", &sgd_step_, "fused momentum-SGD step");
+  m.def("bn_fwd_reduce", &bn_fwd_reduce_py);
+  m.def("bn_fwd_finalize", &bn_fwd_finalize_py);
+  m.def("bn_eval_prep", &bn_eval_prep_py);
+  m.def("bn_fwd_apply", &bn_fwd_apply_py);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce_py);
+  m.def("bn_bwd_apply", &bn_bwd_apply_py);
+  m.def("bn_bwd_apply_eval", &bn_bwd_apply_eval_py);
 }

@@ -1,0 +1,441 @@
+// Fused NHWC BatchNorm(+residual-add)(+ReLU) for gfx950 (CDNA4), bf16
+// activations / fp32 statistics.
+//
+// Replaces, per BN layer per step, the stack of: MIOpen spatial BN
+// (3 kernels fwd + 3 bwd), autocast bf16<->fp32 casts (torch runs BN in
+// fp32 under autocast), the separate ReLU/clamp kernels, the residual
+// add, and the num_batches_tracked counter bump — profiled together at
+// ~38% of a graphed ResNet-50 bs=32 step (profiles/r01_*).
+//
+// Layout: channels_last (NHWC): a [N,C,H,W] tensor is a row-major
+// M x C matrix with M = N*H*W.  Per-channel statistics are column sums.
+//
+// Thread geometry for reductions: blockDim = 256 = TX * TY where
+// TX = C/8 (each thread owns 8 consecutive bf16 channels = one 16-B
+// load) and TY row-parallel lanes; a block strides over rows, partials
+// are reduced through LDS and atomically added to a fp32 scratch[2C].
+// C in {8,16,...,2048} covers every ResNet width.
+//
+// bf16 conversion is round-to-nearest-even, done in-kernel (activations
+// stay bf16 end-to-end; statistics/weights fp32).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#define THREADS 256
+
+namespace {
+
+typedef unsigned short ushort_t;
+
+__device__ __forceinline__ float b2f(ushort_t u) {
+  union { unsigned int i; float f; } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+
+__device__ __forceinline__ ushort_t f2b(float f) {
+  union { unsigned int i; float f; } v;
+  v.f = f;
+  unsigned int r = v.i + 0x7FFFu + ((v.i >> 16) & 1u);
+  return (ushort_t)(r >> 16);
+}
+
+struct U4 { unsigned int x, y, z, w; };  // 8 bf16
+
+__device__ __forceinline__ U4 ld8h(const ushort_t* p) {
+  return *reinterpret_cast<const U4*>(p);
+}
+__device__ __forceinline__ void st8h(ushort_t* p, U4 v) {
+  *reinterpret_cast<U4*>(p) = v;
+}
+
+__device__ __forceinline__ void unpack8(U4 v, float* f) {
+  f[0] = b2f((ushort_t)(v.x & 0xFFFF)); f[1] = b2f((ushort_t)(v.x >> 16));
+  f[2] = b2f((ushort_t)(v.y & 0xFFFF)); f[3] = b2f((ushort_t)(v.y >> 16));
+  f[4] = b2f((ushort_t)(v.z & 0xFFFF)); f[5] = b2f((ushort_t)(v.z >> 16));
+  f[6] = b2f((ushort_t)(v.w & 0xFFFF)); f[7] = b2f((ushort_t)(v.w >> 16));
+}
+
+__device__ __forceinline__ U4 pack8(const float* f) {
+  U4 v;
+  v.x = (unsigned int)f2b(f[0]) | ((unsigned int)f2b(f[1]) << 16);
+  v.y = (unsigned int)f2b(f[2]) | ((unsigned int)f2b(f[3]) << 16);
+  v.z = (unsigned int)f2b(f[4]) | ((unsigned int)f2b(f[5]) << 16);
+  v.w = (unsigned int)f2b(f[6]) | ((unsigned int)f2b(f[7]) << 16);
+  return v;
+}
+
+// ------------------------------------------------------------ fwd reduce
+// scratch[0..C) += column sums of x; scratch[C..2C) += column sums of x^2
+__global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
+                                float* __restrict__ scratch,
+                                int64_t M, int C) {
+  const int TX = C >> 3;                 // threads per row
+  const int tx = threadIdx.x % TX;
+  const int ty = threadIdx.x / TX;
+  const int TY = blockDim.x / TX;
+  const int c0 = tx << 3;
+
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float f[8];
+  const int64_t row0 = (int64_t)blockIdx.x * TY + ty;
+  const int64_t rstride = (int64_t)gridDim.x * TY;
+  for (int64_t r = row0; r < M; r += rstride) {
+    U4 v = ld8h(x + r * C + c0);
+    unpack8(v, f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      s[j] += f[j];
+      q[j] += f[j] * f[j];
+    }
+  }
+
+  // LDS tree-reduce over TY
+  __shared__ float lds[THREADS * 16];
+  float* mine = lds + threadIdx.x * 16;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mine[j] = s[j];
+    mine[8 + j] = q[j];
+  }
+  __syncthreads();
+  if (ty == 0) {
+    for (int t = 1; t < TY; ++t) {
+      const float* other = lds + (t * TX + tx) * 16;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s[j] += other[j];
+        q[j] += other[8 + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(scratch + c0 + j, s[j]);
+      atomicAdd(scratch + C + c0 + j, q[j]);
+    }
+  }
+}
+
+// ---------------------------------------------------------- fwd finalize
+// From raw sums: mean/invstd, running-stat update, and the per-channel
+// scale/shift pair the apply kernel consumes.
+__global__ void k_bn_fwd_finalize(const float* __restrict__ scratch,
+                                  const float* __restrict__ gamma,
+                                  const float* __restrict__ beta,
+                                  float* __restrict__ running_mean,
+                                  float* __restrict__ running_var,
+                                  float* __restrict__ save_mean,
+                                  float* __restrict__ save_invstd,
+                                  float* __restrict__ scale_shift,
+                                  float momentum, float eps, int64_t M,
+                                  int C, bool update_running) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float inv_m = 1.0f / (float)M;
+  const float mean = scratch[c] * inv_m;
+  float var = scratch[C + c] * inv_m - mean * mean;
+  if (var < 0.f) var = 0.f;
+  const float invstd = rsqrtf(var + eps);
+  save_mean[c] = mean;
+  save_invstd[c] = invstd;
+  if (update_running) {
+    const float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] += momentum * (mean - running_mean[c]);
+    running_var[c] += momentum * (unbiased - running_var[c]);
+  }
+  const float sc = gamma[c] * invstd;
+  scale_shift[c] = sc;
+  scale_shift[C + c] = beta[c] - mean * sc;
+}
+
+// ------------------------------------------------------------- eval prep
+// scale/shift straight from running stats (inference path)
+__global__ void k_bn_eval_prep(const float* __restrict__ running_mean,
+                               const float* __restrict__ running_var,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ beta,
+                               float* __restrict__ scale_shift, float eps,
+                               int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float invstd = rsqrtf(running_var[c] + eps);
+  const float sc = gamma[c] * invstd;
+  scale_shift[c] = sc;
+  scale_shift[C + c] = beta[c] - running_mean[c] * sc;
+}
+
+// ------------------------------------------------------------- fwd apply
+// y = [relu]( x*scale + shift [+ res] ), all bf16 in/out
+template <bool kRelu, bool kRes>
+__global__ void k_bn_fwd_apply(const ushort_t* __restrict__ x,
+                               const ushort_t* __restrict__ res,
+                               ushort_t* __restrict__ y,
+                               const float* __restrict__ scale_shift,
+                               int64_t total8, int C) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float f[8], rz[8];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total8; i += stride) {
+    const int64_t e0 = i << 3;
+    const int c0 = (int)(e0 % C);
+    U4 v = ld8h(x + e0);
+    unpack8(v, f);
+    if (kRes) {
+      U4 rv = ld8h(res + e0);
+      unpack8(rv, rz);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float t = f[j] * scale_shift[c0 + j] + scale_shift[C + c0 + j];
+      if (kRes) t += rz[j];
+      if (kRelu) t = t > 0.f ? t : 0.f;
+      f[j] = t;
+    }
+    st8h(y + e0, pack8(f));
+  }
+}
+
+// ------------------------------------------------------------ bwd reduce
+// scratch[0..C) += sum(dy_eff * xhat)  (raw dgamma)
+// scratch[C..2C) += sum(dy_eff)        (raw dbeta)
+// dy_eff = relu? dy * (y > 0) : dy ; xhat = (x - mean) * invstd
+template <bool kRelu>
+__global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
+                                const ushort_t* __restrict__ dy,
+                                const ushort_t* __restrict__ y,
+                                const float* __restrict__ save_mean,
+                                const float* __restrict__ save_invstd,
+                                float* __restrict__ scratch, int64_t M,
+                                int C) {
+  const int TX = C >> 3;
+  const int tx = threadIdx.x % TX;
+  const int ty = threadIdx.x / TX;
+  const int TY = blockDim.x / TX;
+  const int c0 = tx << 3;
+
+  float mean[8], istd[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mean[j] = save_mean[c0 + j];
+    istd[j] = save_invstd[c0 + j];
+  }
+
+  float dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float fx[8], fdy[8], fy[8];
+  const int64_t row0 = (int64_t)blockIdx.x * TY + ty;
+  const int64_t rstride = (int64_t)gridDim.x * TY;
+  for (int64_t r = row0; r < M; r += rstride) {
+    const int64_t off = r * C + c0;
+    unpack8(ld8h(x + off), fx);
+    unpack8(ld8h(dy + off), fdy);
+    if (kRelu) unpack8(ld8h(y + off), fy);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float d = fdy[j];
+      if (kRelu && !(fy[j] > 0.f)) d = 0.f;
+      db[j] += d;
+      dg[j] += d * (fx[j] - mean[j]) * istd[j];
+    }
+  }
+
+  __shared__ float lds[THREADS * 16];
+  float* mine = lds + threadIdx.x * 16;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mine[j] = dg[j];
+    mine[8 + j] = db[j];
+  }
+  __syncthreads();
+  if (ty == 0) {
+    for (int t = 1; t < TY; ++t) {
+      const float* other = lds + (t * TX + tx) * 16;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        dg[j] += other[j];
+        db[j] += other[8 + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(scratch + c0 + j, dg[j]);
+      atomicAdd(scratch + C + c0 + j, db[j]);
+    }
+  }
+}
+
+// ------------------------------------------------------------- bwd apply
+// dx = gamma*invstd * (dy_eff - dbeta/M - xhat*dgamma/M)  [training]
+// dres = dy_eff (when the forward fused a residual add)
+template <bool kRelu, bool kRes>
+__global__ void k_bn_bwd_apply(const ushort_t* __restrict__ x,
+                               const ushort_t* __restrict__ dy,
+                               const ushort_t* __restrict__ y,
+                               ushort_t* __restrict__ dx,
+                               ushort_t* __restrict__ dres,
+                               const float* __restrict__ save_mean,
+                               const float* __restrict__ save_invstd,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ scratch,
+                               int64_t total8, int64_t M, int C) {
+  const float inv_m = 1.0f / (float)M;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float fx[8], fdy[8], fy[8], o[8], orz[8];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total8; i += stride) {
+    const int64_t e0 = i << 3;
+    const int c0 = (int)(e0 % C);
+    unpack8(ld8h(x + e0), fx);
+    unpack8(ld8h(dy + e0), fdy);
+    if (kRelu) unpack8(ld8h(y + e0), fy);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      float d = fdy[j];
+      if (kRelu && !(fy[j] > 0.f)) d = 0.f;
+      if (kRes) orz[j] = d;
+      const float istd = save_invstd[c];
+      const float xhat = (fx[j] - save_mean[c]) * istd;
+      const float k2 = scratch[C + c] * inv_m;  // mean dy_eff
+      const float k3 = scratch[c] * inv_m;      // mean dy_eff*xhat
+      o[j] = gamma[c] * istd * (d - k2 - xhat * k3);
+    }
+    st8h(dx + e0, pack8(o));
+    if (kRes) st8h(dres + e0, pack8(orz));
+  }
+}
+
+// eval-mode backward: dx = gamma*invstd*dy_eff (stats are constants)
+template <bool kRelu, bool kRes>
+__global__ void k_bn_bwd_apply_eval(const ushort_t* __restrict__ dy,
+                                    const ushort_t* __restrict__ y,
+                                    ushort_t* __restrict__ dx,
+                                    ushort_t* __restrict__ dres,
+                                    const float* __restrict__ save_invstd,
+                                    const float* __restrict__ gamma,
+                                    int64_t total8, int C) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float fdy[8], fy[8], o[8], orz[8];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total8; i += stride) {
+    const int64_t e0 = i << 3;
+    const int c0 = (int)(e0 % C);
+    unpack8(ld8h(dy + e0), fdy);
+    if (kRelu) unpack8(ld8h(y + e0), fy);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      float d = fdy[j];
+      if (kRelu && !(fy[j] > 0.f)) d = 0.f;
+      if (kRes) orz[j] = d;
+      o[j] = gamma[c] * save_invstd[c] * d;
+    }
+    st8h(dx + e0, pack8(o));
+    if (kRes) st8h(dres + e0, pack8(orz));
+  }
+}
+
+inline int reduce_grid(int64_t M, int C) {
+  const int TY = THREADS / (C >> 3);
+  int64_t blocks = (M + TY - 1) / TY;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+inline int apply_grid(int64_t total8) {
+  int64_t blocks = (total8 + THREADS - 1) / THREADS;
+  if (blocks > 1048576) blocks = 1048576;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+}  // namespace
+
+extern "C" {
+
+void bn_fwd_reduce(const ushort_t* x, float* scratch, int64_t M, int C,
+                   hipStream_t s) {
+  hipLaunchKernelGGL(k_bn_fwd_reduce, dim3(reduce_grid(M, C)), dim3(THREADS),
+                     0, s, x, scratch, M, C);
+}
+
+void bn_fwd_finalize(const float* scratch, const float* gamma,
+                     const float* beta, float* rmean, float* rvar,
+                     float* smean, float* sinvstd, float* scale_shift,
+                     double momentum, double eps, int64_t M, int C,
+                     bool update_running, hipStream_t s) {
+  const int blocks = (C + THREADS - 1) / THREADS;
+  hipLaunchKernelGGL(k_bn_fwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
+                     scratch, gamma, beta, rmean, rvar, smean, sinvstd,
+                     scale_shift, (float)momentum, (float)eps, M, C,
+                     update_running);
+}
+
+void bn_eval_prep(const float* rmean, const float* rvar, const float* gamma,
+                  const float* beta, float* scale_shift, double eps, int C,
+                  hipStream_t s) {
+  const int blocks = (C + THREADS - 1) / THREADS;
+  hipLaunchKernelGGL(k_bn_eval_prep, dim3(blocks), dim3(THREADS), 0, s, rmean,
+                     rvar, gamma, beta, scale_shift, (float)eps, C);
+}
+
+void bn_fwd_apply(const ushort_t* x, const ushort_t* res, ushort_t* y,
+                  const float* scale_shift, int64_t M, int C, bool relu,
+                  hipStream_t s) {
+  const int64_t total8 = M * C / 8;
+  const dim3 grid(apply_grid(total8));
+#define CASE(R, Z)                                                        \
+  hipLaunchKernelGGL((k_bn_fwd_apply<R, Z>), grid, dim3(THREADS), 0, s, x, \
+                     res, y, scale_shift, total8, C)
+  if (relu) { if (res) CASE(true, true); else CASE(true, false); }
+  else      { if (res) CASE(false, true); else CASE(false, false); }
+#undef CASE
+}
+
+void bn_bwd_reduce(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
+                   const float* smean, const float* sinvstd, float* scratch,
+                   int64_t M, int C, bool relu, hipStream_t s) {
+  const dim3 grid(reduce_grid(M, C));
+  if (relu)
+    hipLaunchKernelGGL((k_bn_bwd_reduce<true>), grid, dim3(THREADS), 0, s, x,
+                       dy, y, smean, sinvstd, scratch, M, C);
+  else
+    hipLaunchKernelGGL((k_bn_bwd_reduce<false>), grid, dim3(THREADS), 0, s, x,
+                       dy, y, smean, sinvstd, scratch, M, C);
+}
+
+void bn_bwd_apply(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
+                  ushort_t* dx, ushort_t* dres, const float* smean,
+                  const float* sinvstd, const float* gamma,
+                  const float* scratch, int64_t M, int C, bool relu,
+                  hipStream_t s) {
+  const int64_t total8 = M * C / 8;
+  const dim3 grid(apply_grid(total8));
+#define CASE(R, Z)                                                         \
+  hipLaunchKernelGGL((k_bn_bwd_apply<R, Z>), grid, dim3(THREADS), 0, s, x, \
+                     dy, y, dx, dres, smean, sinvstd, gamma, scratch,      \
+                     total8, M, C)
+  if (relu) { if (dres) CASE(true, true); else CASE(true, false); }
+  else      { if (dres) CASE(false, true); else CASE(false, false); }
+#undef CASE
+}
+
+void bn_bwd_apply_eval(const ushort_t* dy, const ushort_t* y, ushort_t* dx,
+                       ushort_t* dres, const float* sinvstd,
+                       const float* gamma, int64_t M, int C, bool relu,
+                       hipStream_t s) {
+  const int64_t total8 = M * C / 8;
+  const dim3 grid(apply_grid(total8));
+#define CASE(R, Z)                                                          \
+  hipLaunchKernelGGL((k_bn_bwd_apply_eval<R, Z>), grid, dim3(THREADS), 0, s, \
+                     dy, y, dx, dres, sinvstd, gamma, total8, C)
+  if (relu) { if (dres) CASE(true, true); else CASE(true, false); }
+  else      { if (dres) CASE(false, true); else CASE(false, false); }
+#undef CASE
+}
+
+}  // extern "C"

@@ -1,0 +1,158 @@
+"""Fused NHWC BN(+residual)(+ReLU) kernels vs fp32 PyTorch reference."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from stochastic_gradient_push_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+CL = torch.channels_last
+
+
+def dev():
+    return torch.device("cuda", 0)
+
+
+def make_inputs(N, C, H, W, seed=0, residual=False):
+    torch.manual_seed(seed)
+    x = (
+        torch.randn(N, C, H, W, device=dev())
+        .to(torch.bfloat16)
+        .contiguous(memory_format=CL)
+        .requires_grad_(True)
+    )
+    res = None
+    if residual:
+        res = (
+            torch.randn(N, C, H, W, device=dev())
+            .to(torch.bfloat16)
+            .contiguous(memory_format=CL)
+            .requires_grad_(True)
+        )
+    return x, res
+
+
+def reference(x, res, gamma, beta, rmean, rvar, momentum, eps, training, relu):
+    """fp32 oracle of the same op."""
+    y = F.batch_norm(
+        x.float(), rmean, rvar, gamma, beta, training, momentum, eps
+    )
+    if res is not None:
+        y = y + res.float()
+    if relu:
+        y = torch.relu(y)
+    return y
+
+
+SHAPES = [(4, 64, 16, 16), (2, 256, 14, 14), (3, 8, 7, 7), (2, 2048, 7, 7)]
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+@pytest.mark.parametrize("relu", [False, True])
+@pytest.mark.parametrize("residual", [False, True])
+def test_fused_bn_forward_backward(shape, relu, residual):
+    from stochastic_gradient_push_amd.models.layers import _FusedBNFunction
+
+    N, C, H, W = shape
+    x, res = make_inputs(N, C, H, W, residual=residual)
+    gamma = torch.rand(C, device=dev()) + 0.5
+    beta = torch.randn(C, device=dev())
+    gamma_r = gamma.clone().requires_grad_(True)
+    beta_r = beta.clone().requires_grad_(True)
+    gamma_f = gamma.clone().requires_grad_(True)
+    beta_f = beta.clone().requires_grad_(True)
+
+    rmean_f = torch.zeros(C, device=dev())
+    rvar_f = torch.ones(C, device=dev())
+    rmean_r = torch.zeros(C, device=dev())
+    rvar_r = torch.ones(C, device=dev())
+
+    x_r = x.detach().clone().requires_grad_(True)
+    res_r = res.detach().clone().requires_grad_(True) if residual else None
+
+    y = _FusedBNFunction.apply(
+        x, res, gamma_f, beta_f, rmean_f, rvar_f, 0.1, 1e-5, True, relu
+    )
+    y_ref = reference(
+        x_r, res_r, gamma_r, beta_r, rmean_r, rvar_r, 0.1, 1e-5, True, relu
+    )
+    torch.cuda.synchronize()
+    assert torch.allclose(
+        y.float(), y_ref, atol=5e-2, rtol=5e-2
+    ), f"fwd max err {(y.float() - y_ref).abs().max()}"
+    assert torch.allclose(rmean_f, rmean_r, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(rvar_f, rvar_r, atol=5e-2, rtol=5e-2)
+
+    dy = torch.randn_like(y_ref)
+    y.backward(dy.to(torch.bfloat16).contiguous(memory_format=CL))
+    y_ref.backward(dy)
+    torch.cuda.synchronize()
+
+    M = N * H * W
+    assert torch.allclose(
+        x.grad.float(), x_r.grad, atol=8e-2, rtol=8e-2
+    ), f"dx max err {(x.grad.float() - x_r.grad).abs().max()}"
+    # reductions over M elements: scale tolerance
+    tol = 2e-2 * max(1.0, M ** 0.5)
+    assert torch.allclose(gamma_f.grad, gamma_r.grad, atol=tol), (
+        f"dgamma max err {(gamma_f.grad - gamma_r.grad).abs().max()}"
+    )
+    assert torch.allclose(beta_f.grad, beta_r.grad, atol=tol)
+    if residual:
+        assert torch.allclose(
+            res.grad.float(), res_r.grad, atol=5e-2, rtol=5e-2
+        )
+
+
+def test_fused_bn_eval_mode():
+    from stochastic_gradient_push_amd.models.layers import _FusedBNFunction
+
+    N, C, H, W = 2, 64, 8, 8
+    x, _ = make_inputs(N, C, H, W, seed=3)
+    gamma = torch.rand(C, device=dev()) + 0.5
+    beta = torch.randn(C, device=dev())
+    rmean = torch.randn(C, device=dev()) * 0.1
+    rvar = torch.rand(C, device=dev()) + 0.5
+
+    y = _FusedBNFunction.apply(
+        x, None, gamma.clone(), beta.clone(), rmean.clone(), rvar.clone(),
+        0.1, 1e-5, False, True,
+    )
+    x_r = x.detach().float()
+    y_ref = reference(
+        x_r, None, gamma.clone(), beta.clone(), rmean.clone(), rvar.clone(),
+        0.1, 1e-5, False, True,
+    )
+    torch.cuda.synchronize()
+    assert torch.allclose(y.float(), y_ref, atol=5e-2, rtol=5e-2)
+
+
+def test_fused_resnet_step_close_to_native():
+    """One bf16 training step of resnet18 with fused BN tracks the native
+    implementation."""
+    from stochastic_gradient_push_amd.models import build_resnet
+
+    torch.manual_seed(0)
+    mf = build_resnet("resnet18", num_classes=10, norm="fused").to(dev())
+    torch.manual_seed(0)
+    mn = build_resnet("resnet18", num_classes=10, norm="native").to(dev())
+    mf = mf.to(memory_format=CL)
+    mn = mn.to(memory_format=CL)
+
+    x = torch.randn(4, 3, 64, 64, device=dev()).contiguous(memory_format=CL)
+    y = torch.randint(0, 10, (4,), device=dev())
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        lf = F.cross_entropy(mf(x), y)
+        ln = F.cross_entropy(mn(x), y)
+    lf.backward()
+    ln.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(lf) and torch.isfinite(ln)
+    assert abs(lf.item() - ln.item()) < 0.1, (lf.item(), ln.item())
+    # gradient agreement on the first conv (end of the backward chain)
+    gf = dict(mf.named_parameters())["conv1.weight"].grad
+    gn = dict(mn.named_parameters())["conv1.weight"].grad
+    cos = F.cosine_similarity(gf.flatten(), gn.flatten(), dim=0)
+    assert cos.item() > 0.99, cos.item()
