@@ -81,7 +81,9 @@ class _FusedBNFunction(torch.autograd.Function):
         y = torch.empty_like(x)
         scale_shift = torch.empty(2 * C, device=x.device, dtype=torch.float32)
         if training:
-            scratch = torch.zeros(2 * C, device=x.device, dtype=torch.float32)
+            scratch = torch.zeros(
+                8 * 2 * C, device=x.device, dtype=torch.float32
+            )
             smean = torch.empty(C, device=x.device, dtype=torch.float32)
             sinvstd = torch.empty(C, device=x.device, dtype=torch.float32)
             k.bn_fwd_reduce(x, scratch, M, C)
@@ -111,19 +113,18 @@ class _FusedBNFunction(torch.autograd.Function):
         M, C = ctx.bn_shape
         relu = ctx.bn_relu
         dy = dy.contiguous(memory_format=torch.channels_last)
-        scratch = torch.zeros(2 * C, device=x.device, dtype=torch.float32)
+        scratch = torch.zeros(8 * 2 * C, device=x.device, dtype=torch.float32)
         k.bn_bwd_reduce(x, dy, y if relu else None, smean, sinvstd,
                         scratch, M, C, relu)
+        dgamma = torch.empty(C, device=x.device, dtype=torch.float32)
+        dbeta = torch.empty(C, device=x.device, dtype=torch.float32)
+        coef = torch.empty(3 * C, device=x.device, dtype=torch.float32)
+        k.bn_bwd_finalize(scratch, gamma, smean, sinvstd, dgamma, dbeta,
+                          coef, M, C, ctx.bn_training)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.bn_has_res else None
-        if ctx.bn_training:
-            k.bn_bwd_apply(x, dy, y if relu else None, dx, dres, smean,
-                           sinvstd, gamma, scratch, M, C, relu)
-        else:
-            k.bn_bwd_apply_eval(dy, y if relu else None, dx, dres, sinvstd,
-                                gamma, M, C, relu)
-        return (dx, dres, scratch[:C], scratch[C:], None, None, None, None,
-                None, None)
+        k.bn_bwd_apply(x, dy, y if relu else None, dx, dres, coef, M, C, relu)
+        return (dx, dres, dgamma, dbeta, None, None, None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):

@@ -9,9 +9,9 @@
 typedef unsigned short ushort_t;
 
 extern "C" {
-void bn_fwd_reduce(const ushort_t* x, float* scratch, int64_t M, int C,
+void bn_fwd_reduce(const ushort_t* x, float* shadows, int64_t M, int C,
                    hipStream_t s);
-void bn_fwd_finalize(const float* scratch, const float* gamma,
+void bn_fwd_finalize(const float* shadows, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
                      float* smean, float* sinvstd, float* scale_shift,
                      double momentum, double eps, int64_t M, int C,
@@ -23,17 +23,15 @@ void bn_fwd_apply(const ushort_t* x, const ushort_t* res, ushort_t* y,
                   const float* scale_shift, int64_t M, int C, bool relu,
                   hipStream_t s);
 void bn_bwd_reduce(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
-                   const float* smean, const float* sinvstd, float* scratch,
+                   const float* smean, const float* sinvstd, float* shadows,
                    int64_t M, int C, bool relu, hipStream_t s);
+void bn_bwd_finalize(const float* shadows, const float* gamma,
+                     const float* smean, const float* sinvstd, float* dgamma,
+                     float* dbeta, float* coef, int64_t M, int C,
+                     bool training, hipStream_t s);
 void bn_bwd_apply(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
-                  ushort_t* dx, ushort_t* dres, const float* smean,
-                  const float* sinvstd, const float* gamma,
-                  const float* scratch, int64_t M, int C, bool relu,
-                  hipStream_t s);
-void bn_bwd_apply_eval(const ushort_t* dy, const ushort_t* y, ushort_t* dx,
-                       ushort_t* dres, const float* sinvstd,
-                       const float* gamma, int64_t M, int C, bool relu,
-                       hipStream_t s);
+                  ushort_t* dx, ushort_t* dres, const float* coef, int64_t M,
+                  int C, bool relu, hipStream_t s);
 void sgp_scale(float* x, const float* a, int64_t n, hipStream_t stream);
 void sgp_add_scale(float* x, const float* r, const float* a, int64_t n,
                    hipStream_t stream);
@@ -131,11 +129,11 @@ void check_f32(const torch::Tensor& t, const char* name, int64_t numel) {
   TORCH_CHECK(t.numel() == numel, name, " wrong size");
 }
 
-void bn_fwd_reduce_py(torch::Tensor x, torch::Tensor scratch, int64_t M,
+void bn_fwd_reduce_py(torch::Tensor x, torch::Tensor shadows, int64_t M,
                       int64_t C) {
   check_act(x, "x");
-  check_f32(scratch, "scratch", 2 * C);
-  bn_fwd_reduce(bf16_ptr(x), scratch.data_ptr<float>(), M, (int)C,
+  check_f32(shadows, "shadows", 8 * 2 * C);
+  bn_fwd_reduce(bf16_ptr(x), shadows.data_ptr<float>(), M, (int)C,
                 current_stream(x));
 }
 
@@ -145,7 +143,7 @@ void bn_fwd_finalize_py(torch::Tensor scratch, torch::Tensor gamma,
                         torch::Tensor sinvstd, torch::Tensor scale_shift,
                         double momentum, double eps, int64_t M, int64_t C,
                         bool update_running) {
-  check_f32(scratch, "scratch", 2 * C);
+  check_f32(scratch, "shadows", 8 * 2 * C);
   check_f32(gamma, "gamma", C);
   check_f32(beta, "beta", C);
   check_f32(rmean, "rmean", C);
@@ -192,7 +190,7 @@ void bn_bwd_reduce_py(torch::Tensor x, torch::Tensor dy,
                       int64_t M, int64_t C, bool relu) {
   check_act(x, "x");
   check_act(dy, "dy");
-  check_f32(scratch, "scratch", 2 * C);
+  check_f32(scratch, "shadows", 8 * 2 * C);
   const ushort_t* yp = nullptr;
   if (relu) {
     TORCH_CHECK(y.has_value(), "y required for relu backward");
@@ -204,15 +202,30 @@ void bn_bwd_reduce_py(torch::Tensor x, torch::Tensor dy,
                 (int)C, relu, current_stream(x));
 }
 
+void bn_bwd_finalize_py(torch::Tensor shadows, torch::Tensor gamma,
+                        torch::Tensor smean, torch::Tensor sinvstd,
+                        torch::Tensor dgamma, torch::Tensor dbeta,
+                        torch::Tensor coef, int64_t M, int64_t C,
+                        bool training) {
+  check_f32(shadows, "shadows", 8 * 2 * C);
+  check_f32(dgamma, "dgamma", C);
+  check_f32(dbeta, "dbeta", C);
+  check_f32(coef, "coef", 3 * C);
+  bn_bwd_finalize(shadows.data_ptr<float>(), gamma.data_ptr<float>(),
+                  smean.data_ptr<float>(), sinvstd.data_ptr<float>(),
+                  dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                  coef.data_ptr<float>(), M, (int)C, training,
+                  current_stream(shadows));
+}
+
 void bn_bwd_apply_py(torch::Tensor x, torch::Tensor dy,
                      torch::optional<torch::Tensor> y, torch::Tensor dx,
-                     torch::optional<torch::Tensor> dres,
-                     torch::Tensor smean, torch::Tensor sinvstd,
-                     torch::Tensor gamma, torch::Tensor scratch, int64_t M,
-                     int64_t C, bool relu) {
+                     torch::optional<torch::Tensor> dres, torch::Tensor coef,
+                     int64_t M, int64_t C, bool relu) {
   check_act(x, "x");
   check_act(dy, "dy");
   check_act(dx, "dx");
+  check_f32(coef, "coef", 3 * C);
   const ushort_t* yp = nullptr;
   ushort_t* drp = nullptr;
   if (relu) {
@@ -221,28 +234,7 @@ void bn_bwd_apply_py(torch::Tensor x, torch::Tensor dy,
   }
   if (dres.has_value()) drp = bf16_mut(dres.value());
   bn_bwd_apply(bf16_ptr(x), bf16_ptr(dy), yp, bf16_mut(dx), drp,
-               smean.data_ptr<float>(), sinvstd.data_ptr<float>(),
-               gamma.data_ptr<float>(), scratch.data_ptr<float>(), M,
-               (int)C, relu, current_stream(x));
-}
-
-void bn_bwd_apply_eval_py(torch::Tensor dy, torch::optional<torch::Tensor> y,
-                          torch::Tensor dx,
-                          torch::optional<torch::Tensor> dres,
-                          torch::Tensor sinvstd, torch::Tensor gamma,
-                          int64_t M, int64_t C, bool relu) {
-  check_act(dy, "dy");
-  check_act(dx, "dx");
-  const ushort_t* yp = nullptr;
-  ushort_t* drp = nullptr;
-  if (relu) {
-    TORCH_CHECK(y.has_value(), "y required for relu backward");
-    yp = bf16_ptr(y.value());
-  }
-  if (dres.has_value()) drp = bf16_mut(dres.value());
-  bn_bwd_apply_eval(bf16_ptr(dy), yp, bf16_mut(dx), drp,
-                    sinvstd.data_ptr<float>(), gamma.data_ptr<float>(), M,
-                    (int)C, relu, current_stream(dy));
+               coef.data_ptr<float>(), M, (int)C, relu, current_stream(x));
 }
 
 }  // namespace
@@ -259,6 +251,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_eval_prep", &bn_eval_prep_py);
   m.def("bn_fwd_apply", &bn_fwd_apply_py);
   m.def("bn_bwd_reduce", &bn_bwd_reduce_py);
+  m.def("bn_bwd_finalize", &bn_bwd_finalize_py);
   m.def("bn_bwd_apply", &bn_bwd_apply_py);
-  m.def("bn_bwd_apply_eval", &bn_bwd_apply_eval_py);
 }

@@ -10,20 +10,27 @@
 // Layout: channels_last (NHWC): a [N,C,H,W] tensor is a row-major
 // M x C matrix with M = N*H*W.  Per-channel statistics are column sums.
 //
-// Thread geometry for reductions: blockDim = 256 = TX * TY where
-// TX = C/8 (each thread owns 8 consecutive bf16 channels = one 16-B
-// load) and TY row-parallel lanes; a block strides over rows, partials
-// are reduced through LDS and atomically added to a fp32 scratch[2C].
-// C in {8,16,...,2048} covers every ResNet width.
-//
-// bf16 conversion is round-to-nearest-even, done in-kernel (activations
-// stay bf16 end-to-end; statistics/weights fp32).
+// Performance notes (measured on MI355X):
+// * Reductions atomically accumulate into NSHADOW=8 shadow copies of the
+//   per-channel sums, indexed blockIdx%8 — the dispatcher places block b
+//   on XCD b%8, so each shadow stays in one XCD's L2 and cross-block
+//   atomic serialization drops 8x.  A single-copy version measured 400us
+//   on a 100k x 256 reduce (50x off the HBM bound) from exactly this
+//   contention.
+// * Apply kernels exploit that the grid stride (gridDim*256*8 elements)
+//   is always a multiple of C (C = 8*2^k <= 2048 divides 2048), so each
+//   lane's 8-channel group is loop-invariant: all per-channel constants
+//   are hoisted into registers and the inner loop is pure streaming
+//   (one 16-B load [+1 per extra stream], one 16-B store).
+// * bf16 conversion is round-to-nearest-even, in-register; activations
+//   stay bf16 end-to-end (no autocast fp32 round trip).
 
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
 
 #define THREADS 256
+#define NSHADOW 8
 
 namespace {
 
@@ -68,24 +75,21 @@ __device__ __forceinline__ U4 pack8(const float* f) {
 }
 
 // ------------------------------------------------------------ fwd reduce
-// scratch[0..C) += column sums of x; scratch[C..2C) += column sums of x^2
+// shadow[blockIdx%8][0..C) += col-sums of x ; [C..2C) += col-sums of x^2
 __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
-                                float* __restrict__ scratch,
+                                float* __restrict__ shadows,
                                 int64_t M, int C) {
-  const int TX = C >> 3;                 // threads per row
+  const int TX = C >> 3;
   const int tx = threadIdx.x % TX;
   const int ty = threadIdx.x / TX;
   const int TY = blockDim.x / TX;
   const int c0 = tx << 3;
 
-  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  float f[8];
+  float s[8] = {0}, q[8] = {0}, f[8];
   const int64_t row0 = (int64_t)blockIdx.x * TY + ty;
   const int64_t rstride = (int64_t)gridDim.x * TY;
   for (int64_t r = row0; r < M; r += rstride) {
-    U4 v = ld8h(x + r * C + c0);
-    unpack8(v, f);
+    unpack8(ld8h(x + r * C + c0), f);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       s[j] += f[j];
@@ -93,7 +97,6 @@ __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
     }
   }
 
-  // LDS tree-reduce over TY
   __shared__ float lds[THREADS * 16];
   float* mine = lds + threadIdx.x * 16;
 #pragma unroll
@@ -111,18 +114,18 @@ __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
         q[j] += other[8 + j];
       }
     }
+    float* shadow = shadows + (blockIdx.x % NSHADOW) * 2 * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(scratch + c0 + j, s[j]);
-      atomicAdd(scratch + C + c0 + j, q[j]);
+      atomicAdd(shadow + c0 + j, s[j]);
+      atomicAdd(shadow + C + c0 + j, q[j]);
     }
   }
 }
 
 // ---------------------------------------------------------- fwd finalize
-// From raw sums: mean/invstd, running-stat update, and the per-channel
-// scale/shift pair the apply kernel consumes.
-__global__ void k_bn_fwd_finalize(const float* __restrict__ scratch,
+// Collapse shadows; mean/invstd; running-stat update; scale/shift pair.
+__global__ void k_bn_fwd_finalize(const float* __restrict__ shadows,
                                   const float* __restrict__ gamma,
                                   const float* __restrict__ beta,
                                   float* __restrict__ running_mean,
@@ -134,9 +137,15 @@ __global__ void k_bn_fwd_finalize(const float* __restrict__ scratch,
                                   int C, bool update_running) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float s = 0.f, q = 0.f;
+#pragma unroll
+  for (int k = 0; k < NSHADOW; ++k) {
+    s += shadows[k * 2 * C + c];
+    q += shadows[k * 2 * C + C + c];
+  }
   const float inv_m = 1.0f / (float)M;
-  const float mean = scratch[c] * inv_m;
-  float var = scratch[C + c] * inv_m - mean * mean;
+  const float mean = s * inv_m;
+  float var = q * inv_m - mean * mean;
   if (var < 0.f) var = 0.f;
   const float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
@@ -152,7 +161,6 @@ __global__ void k_bn_fwd_finalize(const float* __restrict__ scratch,
 }
 
 // ------------------------------------------------------------- eval prep
-// scale/shift straight from running stats (inference path)
 __global__ void k_bn_eval_prep(const float* __restrict__ running_mean,
                                const float* __restrict__ running_var,
                                const float* __restrict__ gamma,
@@ -168,28 +176,31 @@ __global__ void k_bn_eval_prep(const float* __restrict__ running_mean,
 }
 
 // ------------------------------------------------------------- fwd apply
-// y = [relu]( x*scale + shift [+ res] ), all bf16 in/out
+// y = [relu]( x*scale + shift [+ res] ) — per-channel constants hoisted
 template <bool kRelu, bool kRes>
 __global__ void k_bn_fwd_apply(const ushort_t* __restrict__ x,
                                const ushort_t* __restrict__ res,
                                ushort_t* __restrict__ y,
                                const float* __restrict__ scale_shift,
                                int64_t total8, int C) {
+  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  // stride*8 % C == 0 -> channel group is loop-invariant
+  const int c0 = (int)((i0 << 3) % C);
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale_shift[c0 + j];
+    sh[j] = scale_shift[C + c0 + j];
+  }
   float f[8], rz[8];
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total8; i += stride) {
+  for (int64_t i = i0; i < total8; i += stride) {
     const int64_t e0 = i << 3;
-    const int c0 = (int)(e0 % C);
-    U4 v = ld8h(x + e0);
-    unpack8(v, f);
-    if (kRes) {
-      U4 rv = ld8h(res + e0);
-      unpack8(rv, rz);
-    }
+    unpack8(ld8h(x + e0), f);
+    if (kRes) unpack8(ld8h(res + e0), rz);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float t = f[j] * scale_shift[c0 + j] + scale_shift[C + c0 + j];
+      float t = f[j] * sc[j] + sh[j];
       if (kRes) t += rz[j];
       if (kRelu) t = t > 0.f ? t : 0.f;
       f[j] = t;
@@ -199,16 +210,14 @@ __global__ void k_bn_fwd_apply(const ushort_t* __restrict__ x,
 }
 
 // ------------------------------------------------------------ bwd reduce
-// scratch[0..C) += sum(dy_eff * xhat)  (raw dgamma)
-// scratch[C..2C) += sum(dy_eff)        (raw dbeta)
-// dy_eff = relu? dy * (y > 0) : dy ; xhat = (x - mean) * invstd
+// shadow[b%8][0..C) += sum(dy_eff * xhat) ; [C..2C) += sum(dy_eff)
 template <bool kRelu>
 __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
                                 const ushort_t* __restrict__ dy,
                                 const ushort_t* __restrict__ y,
                                 const float* __restrict__ save_mean,
                                 const float* __restrict__ save_invstd,
-                                float* __restrict__ scratch, int64_t M,
+                                float* __restrict__ shadows, int64_t M,
                                 int C) {
   const int TX = C >> 3;
   const int tx = threadIdx.x % TX;
@@ -223,9 +232,7 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
     istd[j] = save_invstd[c0 + j];
   }
 
-  float dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  float fx[8], fdy[8], fy[8];
+  float dg[8] = {0}, db[8] = {0}, fx[8], fdy[8], fy[8];
   const int64_t row0 = (int64_t)blockIdx.x * TY + ty;
   const int64_t rstride = (int64_t)gridDim.x * TY;
   for (int64_t r = row0; r < M; r += rstride) {
@@ -259,79 +266,85 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
         db[j] += other[8 + j];
       }
     }
+    float* shadow = shadows + (blockIdx.x % NSHADOW) * 2 * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(scratch + c0 + j, dg[j]);
-      atomicAdd(scratch + C + c0 + j, db[j]);
+      atomicAdd(shadow + c0 + j, dg[j]);
+      atomicAdd(shadow + C + c0 + j, db[j]);
     }
   }
 }
 
+// ---------------------------------------------------------- bwd finalize
+// Collapse shadows -> dgamma/dbeta, and the 3 per-channel coefficients of
+// the affine form dx = A*dy_eff + D*x + B:
+//   A = gamma*invstd
+//   D = -invstd*A*dgamma/M
+//   B = -A*dbeta/M + mean*invstd*A*dgamma/M  ( = -A*k2 - D*(-mean)... )
+// (training; eval uses A only with D=B=0)
+__global__ void k_bn_bwd_finalize(const float* __restrict__ shadows,
+                                  const float* __restrict__ gamma,
+                                  const float* __restrict__ save_mean,
+                                  const float* __restrict__ save_invstd,
+                                  float* __restrict__ dgamma,
+                                  float* __restrict__ dbeta,
+                                  float* __restrict__ coef, int64_t M,
+                                  int C, bool training) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float dg = 0.f, db = 0.f;
+#pragma unroll
+  for (int k = 0; k < NSHADOW; ++k) {
+    dg += shadows[k * 2 * C + c];
+    db += shadows[k * 2 * C + C + c];
+  }
+  dgamma[c] = dg;
+  dbeta[c] = db;
+  const float istd = save_invstd[c];
+  const float A = gamma[c] * istd;
+  float D = 0.f, B = 0.f;
+  if (training) {
+    const float inv_m = 1.0f / (float)M;
+    D = -istd * A * dg * inv_m;
+    B = -A * db * inv_m - D * save_mean[c];
+  }
+  coef[c] = A;
+  coef[C + c] = D;
+  coef[2 * C + c] = B;
+}
+
 // ------------------------------------------------------------- bwd apply
-// dx = gamma*invstd * (dy_eff - dbeta/M - xhat*dgamma/M)  [training]
-// dres = dy_eff (when the forward fused a residual add)
+// dx = A*dy_eff + D*x + B ; dres = dy_eff — constants hoisted per lane
 template <bool kRelu, bool kRes>
 __global__ void k_bn_bwd_apply(const ushort_t* __restrict__ x,
                                const ushort_t* __restrict__ dy,
                                const ushort_t* __restrict__ y,
                                ushort_t* __restrict__ dx,
                                ushort_t* __restrict__ dres,
-                               const float* __restrict__ save_mean,
-                               const float* __restrict__ save_invstd,
-                               const float* __restrict__ gamma,
-                               const float* __restrict__ scratch,
-                               int64_t total8, int64_t M, int C) {
-  const float inv_m = 1.0f / (float)M;
+                               const float* __restrict__ coef,
+                               int64_t total8, int C) {
+  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int c0 = (int)((i0 << 3) % C);
+  float A[8], D[8], B[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    A[j] = coef[c0 + j];
+    D[j] = coef[C + c0 + j];
+    B[j] = coef[2 * C + c0 + j];
+  }
   float fx[8], fdy[8], fy[8], o[8], orz[8];
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total8; i += stride) {
+  for (int64_t i = i0; i < total8; i += stride) {
     const int64_t e0 = i << 3;
-    const int c0 = (int)(e0 % C);
     unpack8(ld8h(x + e0), fx);
     unpack8(ld8h(dy + e0), fdy);
     if (kRelu) unpack8(ld8h(y + e0), fy);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c0 + j;
       float d = fdy[j];
       if (kRelu && !(fy[j] > 0.f)) d = 0.f;
       if (kRes) orz[j] = d;
-      const float istd = save_invstd[c];
-      const float xhat = (fx[j] - save_mean[c]) * istd;
-      const float k2 = scratch[C + c] * inv_m;  // mean dy_eff
-      const float k3 = scratch[c] * inv_m;      // mean dy_eff*xhat
-      o[j] = gamma[c] * istd * (d - k2 - xhat * k3);
-    }
-    st8h(dx + e0, pack8(o));
-    if (kRes) st8h(dres + e0, pack8(orz));
-  }
-}
-
-// eval-mode backward: dx = gamma*invstd*dy_eff (stats are constants)
-template <bool kRelu, bool kRes>
-__global__ void k_bn_bwd_apply_eval(const ushort_t* __restrict__ dy,
-                                    const ushort_t* __restrict__ y,
-                                    ushort_t* __restrict__ dx,
-                                    ushort_t* __restrict__ dres,
-                                    const float* __restrict__ save_invstd,
-                                    const float* __restrict__ gamma,
-                                    int64_t total8, int C) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  float fdy[8], fy[8], o[8], orz[8];
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total8; i += stride) {
-    const int64_t e0 = i << 3;
-    const int c0 = (int)(e0 % C);
-    unpack8(ld8h(dy + e0), fdy);
-    if (kRelu) unpack8(ld8h(y + e0), fy);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int c = c0 + j;
-      float d = fdy[j];
-      if (kRelu && !(fy[j] > 0.f)) d = 0.f;
-      if (kRes) orz[j] = d;
-      o[j] = gamma[c] * save_invstd[c] * d;
+      o[j] = A[j] * d + D[j] * fx[j] + B[j];
     }
     st8h(dx + e0, pack8(o));
     if (kRes) st8h(dres + e0, pack8(orz));
@@ -341,14 +354,14 @@ __global__ void k_bn_bwd_apply_eval(const ushort_t* __restrict__ dy,
 inline int reduce_grid(int64_t M, int C) {
   const int TY = THREADS / (C >> 3);
   int64_t blocks = (M + TY - 1) / TY;
-  if (blocks > 2048) blocks = 2048;
+  if (blocks > 1024) blocks = 1024;
   if (blocks < 1) blocks = 1;
   return (int)blocks;
 }
 
 inline int apply_grid(int64_t total8) {
   int64_t blocks = (total8 + THREADS - 1) / THREADS;
-  if (blocks > 1048576) blocks = 1048576;
+  if (blocks > 4096) blocks = 4096;
   if (blocks < 1) blocks = 1;
   return (int)blocks;
 }
@@ -357,20 +370,20 @@ inline int apply_grid(int64_t total8) {
 
 extern "C" {
 
-void bn_fwd_reduce(const ushort_t* x, float* scratch, int64_t M, int C,
+void bn_fwd_reduce(const ushort_t* x, float* shadows, int64_t M, int C,
                    hipStream_t s) {
   hipLaunchKernelGGL(k_bn_fwd_reduce, dim3(reduce_grid(M, C)), dim3(THREADS),
-                     0, s, x, scratch, M, C);
+                     0, s, x, shadows, M, C);
 }
 
-void bn_fwd_finalize(const float* scratch, const float* gamma,
+void bn_fwd_finalize(const float* shadows, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
                      float* smean, float* sinvstd, float* scale_shift,
                      double momentum, double eps, int64_t M, int C,
                      bool update_running, hipStream_t s) {
   const int blocks = (C + THREADS - 1) / THREADS;
   hipLaunchKernelGGL(k_bn_fwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
-                     scratch, gamma, beta, rmean, rvar, smean, sinvstd,
+                     shadows, gamma, beta, rmean, rvar, smean, sinvstd,
                      scale_shift, (float)momentum, (float)eps, M, C,
                      update_running);
 }
@@ -397,42 +410,35 @@ void bn_fwd_apply(const ushort_t* x, const ushort_t* res, ushort_t* y,
 }
 
 void bn_bwd_reduce(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
-                   const float* smean, const float* sinvstd, float* scratch,
+                   const float* smean, const float* sinvstd, float* shadows,
                    int64_t M, int C, bool relu, hipStream_t s) {
   const dim3 grid(reduce_grid(M, C));
   if (relu)
     hipLaunchKernelGGL((k_bn_bwd_reduce<true>), grid, dim3(THREADS), 0, s, x,
-                       dy, y, smean, sinvstd, scratch, M, C);
+                       dy, y, smean, sinvstd, shadows, M, C);
   else
     hipLaunchKernelGGL((k_bn_bwd_reduce<false>), grid, dim3(THREADS), 0, s, x,
-                       dy, y, smean, sinvstd, scratch, M, C);
+                       dy, y, smean, sinvstd, shadows, M, C);
+}
+
+void bn_bwd_finalize(const float* shadows, const float* gamma,
+                     const float* smean, const float* sinvstd, float* dgamma,
+                     float* dbeta, float* coef, int64_t M, int C,
+                     bool training, hipStream_t s) {
+  const int blocks = (C + THREADS - 1) / THREADS;
+  hipLaunchKernelGGL(k_bn_bwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
+                     shadows, gamma, smean, sinvstd, dgamma, dbeta, coef, M,
+                     C, training);
 }
 
 void bn_bwd_apply(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
-                  ushort_t* dx, ushort_t* dres, const float* smean,
-                  const float* sinvstd, const float* gamma,
-                  const float* scratch, int64_t M, int C, bool relu,
-                  hipStream_t s) {
+                  ushort_t* dx, ushort_t* dres, const float* coef, int64_t M,
+                  int C, bool relu, hipStream_t s) {
   const int64_t total8 = M * C / 8;
   const dim3 grid(apply_grid(total8));
 #define CASE(R, Z)                                                         \
   hipLaunchKernelGGL((k_bn_bwd_apply<R, Z>), grid, dim3(THREADS), 0, s, x, \
-                     dy, y, dx, dres, smean, sinvstd, gamma, scratch,      \
-                     total8, M, C)
-  if (relu) { if (dres) CASE(true, true); else CASE(true, false); }
-  else      { if (dres) CASE(false, true); else CASE(false, false); }
-#undef CASE
-}
-
-void bn_bwd_apply_eval(const ushort_t* dy, const ushort_t* y, ushort_t* dx,
-                       ushort_t* dres, const float* sinvstd,
-                       const float* gamma, int64_t M, int C, bool relu,
-                       hipStream_t s) {
-  const int64_t total8 = M * C / 8;
-  const dim3 grid(apply_grid(total8));
-#define CASE(R, Z)                                                          \
-  hipLaunchKernelGGL((k_bn_bwd_apply_eval<R, Z>), grid, dim3(THREADS), 0, s, \
-                     dy, y, dx, dres, sinvstd, gamma, total8, C)
+                     dy, y, dx, dres, coef, total8, C)
   if (relu) { if (dres) CASE(true, true); else CASE(true, false); }
   else      { if (dres) CASE(false, true); else CASE(false, false); }
 #undef CASE

@@ -92,8 +92,8 @@ def test_fused_bn_forward_backward(shape, relu, residual):
 
     M = N * H * W
     assert torch.allclose(
-        x.grad.float(), x_r.grad, atol=8e-2, rtol=8e-2
-    ), f"dx max err {(x.grad.float() - x_r.grad).abs().max()}"
+        x.grad.float(), x_r.grad.float(), atol=8e-2, rtol=8e-2
+    ), f"dx max err {(x.grad.float() - x_r.grad.float()).abs().max()}"
     # reductions over M elements: scale tolerance
     tol = 2e-2 * max(1.0, M ** 0.5)
     assert torch.allclose(gamma_f.grad, gamma_r.grad, atol=tol), (
@@ -102,7 +102,7 @@ def test_fused_bn_forward_backward(shape, relu, residual):
     assert torch.allclose(beta_f.grad, beta_r.grad, atol=tol)
     if residual:
         assert torch.allclose(
-            res.grad.float(), res_r.grad, atol=5e-2, rtol=5e-2
+            res.grad.float(), res_r.grad.float(), atol=5e-2, rtol=5e-2
         )
 
 
