@@ -81,8 +81,9 @@ class _FusedBNFunction(torch.autograd.Function):
         y = torch.empty_like(x)
         scale_shift = torch.empty(2 * C, device=x.device, dtype=torch.float32)
         if training:
-            scratch = torch.zeros(
-                8 * 2 * C, device=x.device, dtype=torch.float32
+            scratch = torch.empty(
+                k.bn_partials_numel(M, C), device=x.device,
+                dtype=torch.float32,
             )
             smean = torch.empty(C, device=x.device, dtype=torch.float32)
             sinvstd = torch.empty(C, device=x.device, dtype=torch.float32)
@@ -113,7 +114,9 @@ class _FusedBNFunction(torch.autograd.Function):
         M, C = ctx.bn_shape
         relu = ctx.bn_relu
         dy = dy.contiguous(memory_format=torch.channels_last)
-        scratch = torch.zeros(8 * 2 * C, device=x.device, dtype=torch.float32)
+        scratch = torch.empty(
+            k.bn_partials_numel(M, C), device=x.device, dtype=torch.float32
+        )
         k.bn_bwd_reduce(x, dy, y if relu else None, smean, sinvstd,
                         scratch, M, C, relu)
         dgamma = torch.empty(C, device=x.device, dtype=torch.float32)

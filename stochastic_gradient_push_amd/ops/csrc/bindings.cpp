@@ -9,7 +9,8 @@
 typedef unsigned short ushort_t;
 
 extern "C" {
-void bn_fwd_reduce(const ushort_t* x, float* shadows, int64_t M, int C,
+int bn_reduce_nblocks(int64_t M, int C);
+void bn_fwd_reduce(const ushort_t* x, float* partials, int64_t M, int C,
                    hipStream_t s);
 void bn_fwd_finalize(const float* shadows, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
@@ -129,11 +130,15 @@ void check_f32(const torch::Tensor& t, const char* name, int64_t numel) {
   TORCH_CHECK(t.numel() == numel, name, " wrong size");
 }
 
-void bn_fwd_reduce_py(torch::Tensor x, torch::Tensor shadows, int64_t M,
+int64_t bn_partials_numel(int64_t M, int64_t C) {
+  return (int64_t)bn_reduce_nblocks(M, (int)C) * 2 * C;
+}
+
+void bn_fwd_reduce_py(torch::Tensor x, torch::Tensor partials, int64_t M,
                       int64_t C) {
   check_act(x, "x");
-  check_f32(shadows, "shadows", 8 * 2 * C);
-  bn_fwd_reduce(bf16_ptr(x), shadows.data_ptr<float>(), M, (int)C,
+  check_f32(partials, "partials", bn_partials_numel(M, C));
+  bn_fwd_reduce(bf16_ptr(x), partials.data_ptr<float>(), M, (int)C,
                 current_stream(x));
 }
 
@@ -143,7 +148,7 @@ void bn_fwd_finalize_py(torch::Tensor scratch, torch::Tensor gamma,
                         torch::Tensor sinvstd, torch::Tensor scale_shift,
                         double momentum, double eps, int64_t M, int64_t C,
                         bool update_running) {
-  check_f32(scratch, "shadows", 8 * 2 * C);
+  check_f32(scratch, "partials", bn_partials_numel(M, C));
   check_f32(gamma, "gamma", C);
   check_f32(beta, "beta", C);
   check_f32(rmean, "rmean", C);
@@ -190,7 +195,7 @@ void bn_bwd_reduce_py(torch::Tensor x, torch::Tensor dy,
                       int64_t M, int64_t C, bool relu) {
   check_act(x, "x");
   check_act(dy, "dy");
-  check_f32(scratch, "shadows", 8 * 2 * C);
+  check_f32(scratch, "partials", bn_partials_numel(M, C));
   const ushort_t* yp = nullptr;
   if (relu) {
     TORCH_CHECK(y.has_value(), "y required for relu backward");
@@ -202,20 +207,20 @@ void bn_bwd_reduce_py(torch::Tensor x, torch::Tensor dy,
                 (int)C, relu, current_stream(x));
 }
 
-void bn_bwd_finalize_py(torch::Tensor shadows, torch::Tensor gamma,
+void bn_bwd_finalize_py(torch::Tensor partials, torch::Tensor gamma,
                         torch::Tensor smean, torch::Tensor sinvstd,
                         torch::Tensor dgamma, torch::Tensor dbeta,
                         torch::Tensor coef, int64_t M, int64_t C,
                         bool training) {
-  check_f32(shadows, "shadows", 8 * 2 * C);
+  check_f32(partials, "partials", bn_partials_numel(M, C));
   check_f32(dgamma, "dgamma", C);
   check_f32(dbeta, "dbeta", C);
   check_f32(coef, "coef", 3 * C);
-  bn_bwd_finalize(shadows.data_ptr<float>(), gamma.data_ptr<float>(),
+  bn_bwd_finalize(partials.data_ptr<float>(), gamma.data_ptr<float>(),
                   smean.data_ptr<float>(), sinvstd.data_ptr<float>(),
                   dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                   coef.data_ptr<float>(), M, (int)C, training,
-                  current_stream(shadows));
+                  current_stream(partials));
 }
 
 void bn_bwd_apply_py(torch::Tensor x, torch::Tensor dy,
@@ -246,6 +251,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_mix_", &pack_mix_, "x *= a; out = x");
   m.def("average_", &average_, "x = (x + y) / 2");
   m.def("sgd_step_", &sgd_step_, "fused momentum-SGD step");
+  m.def("bn_partials_numel", &bn_partials_numel);
   m.def("bn_fwd_reduce", &bn_fwd_reduce_py);
   m.def("bn_fwd_finalize", &bn_fwd_finalize_py);
   m.def("bn_eval_prep", &bn_eval_prep_py);

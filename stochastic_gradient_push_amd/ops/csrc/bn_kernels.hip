@@ -11,12 +11,13 @@
 // M x C matrix with M = N*H*W.  Per-channel statistics are column sums.
 //
 // Performance notes (measured on MI355X):
-// * Reductions atomically accumulate into NSHADOW=8 shadow copies of the
-//   per-channel sums, indexed blockIdx%8 — the dispatcher places block b
-//   on XCD b%8, so each shadow stays in one XCD's L2 and cross-block
-//   atomic serialization drops 8x.  A single-copy version measured 400us
-//   on a 100k x 256 reduce (50x off the HBM bound) from exactly this
-//   contention.
+// * Reductions are two-stage and atomic-free: each block stores its
+//   per-channel partial sums to partials[block][2C] (plain stores), and
+//   the finalize kernel sums the <=512 partials.  A single-copy
+//   atomicAdd version measured 400us on a 100k x 256 reduce (50x off the
+//   HBM bound) from cross-block serialization on 2C addresses; an
+//   8-shadow version still measured 44us.  Partial-store is also
+//   deterministic (fixed summation order).
 // * Apply kernels exploit that the grid stride (gridDim*256*8 elements)
 //   is always a multiple of C (C = 8*2^k <= 2048 divides 2048), so each
 //   lane's 8-channel group is loop-invariant: all per-channel constants
@@ -30,7 +31,6 @@
 #include <cstdint>
 
 #define THREADS 256
-#define NSHADOW 8
 
 namespace {
 
@@ -77,7 +77,7 @@ __device__ __forceinline__ U4 pack8(const float* f) {
 // ------------------------------------------------------------ fwd reduce
 // shadow[blockIdx%8][0..C) += col-sums of x ; [C..2C) += col-sums of x^2
 __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
-                                float* __restrict__ shadows,
+                                float* __restrict__ partials,
                                 int64_t M, int C) {
   const int TX = C >> 3;
   const int tx = threadIdx.x % TX;
@@ -114,18 +114,18 @@ __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
         q[j] += other[8 + j];
       }
     }
-    float* shadow = shadows + (blockIdx.x % NSHADOW) * 2 * C;
+    float* part = partials + (int64_t)blockIdx.x * 2 * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(shadow + c0 + j, s[j]);
-      atomicAdd(shadow + C + c0 + j, q[j]);
+      part[c0 + j] = s[j];
+      part[C + c0 + j] = q[j];
     }
   }
 }
 
 // ---------------------------------------------------------- fwd finalize
 // Collapse shadows; mean/invstd; running-stat update; scale/shift pair.
-__global__ void k_bn_fwd_finalize(const float* __restrict__ shadows,
+__global__ void k_bn_fwd_finalize(const float* __restrict__ partials,
                                   const float* __restrict__ gamma,
                                   const float* __restrict__ beta,
                                   float* __restrict__ running_mean,
@@ -134,14 +134,13 @@ __global__ void k_bn_fwd_finalize(const float* __restrict__ shadows,
                                   float* __restrict__ save_invstd,
                                   float* __restrict__ scale_shift,
                                   float momentum, float eps, int64_t M,
-                                  int C, bool update_running) {
+                                  int C, int nblocks, bool update_running) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float s = 0.f, q = 0.f;
-#pragma unroll
-  for (int k = 0; k < NSHADOW; ++k) {
-    s += shadows[k * 2 * C + c];
-    q += shadows[k * 2 * C + C + c];
+  for (int k = 0; k < nblocks; ++k) {
+    s += partials[(int64_t)k * 2 * C + c];
+    q += partials[(int64_t)k * 2 * C + C + c];
   }
   const float inv_m = 1.0f / (float)M;
   const float mean = s * inv_m;
@@ -217,7 +216,7 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
                                 const ushort_t* __restrict__ y,
                                 const float* __restrict__ save_mean,
                                 const float* __restrict__ save_invstd,
-                                float* __restrict__ shadows, int64_t M,
+                                float* __restrict__ partials, int64_t M,
                                 int C) {
   const int TX = C >> 3;
   const int tx = threadIdx.x % TX;
@@ -266,11 +265,11 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
         db[j] += other[8 + j];
       }
     }
-    float* shadow = shadows + (blockIdx.x % NSHADOW) * 2 * C;
+    float* part = partials + (int64_t)blockIdx.x * 2 * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(shadow + c0 + j, dg[j]);
-      atomicAdd(shadow + C + c0 + j, db[j]);
+      part[c0 + j] = dg[j];
+      part[C + c0 + j] = db[j];
     }
   }
 }
@@ -282,21 +281,20 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
 //   D = -invstd*A*dgamma/M
 //   B = -A*dbeta/M + mean*invstd*A*dgamma/M  ( = -A*k2 - D*(-mean)... )
 // (training; eval uses A only with D=B=0)
-__global__ void k_bn_bwd_finalize(const float* __restrict__ shadows,
+__global__ void k_bn_bwd_finalize(const float* __restrict__ partials,
                                   const float* __restrict__ gamma,
                                   const float* __restrict__ save_mean,
                                   const float* __restrict__ save_invstd,
                                   float* __restrict__ dgamma,
                                   float* __restrict__ dbeta,
                                   float* __restrict__ coef, int64_t M,
-                                  int C, bool training) {
+                                  int C, int nblocks, bool training) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float dg = 0.f, db = 0.f;
-#pragma unroll
-  for (int k = 0; k < NSHADOW; ++k) {
-    dg += shadows[k * 2 * C + c];
-    db += shadows[k * 2 * C + C + c];
+  for (int k = 0; k < nblocks; ++k) {
+    dg += partials[(int64_t)k * 2 * C + c];
+    db += partials[(int64_t)k * 2 * C + C + c];
   }
   dgamma[c] = dg;
   dbeta[c] = db;
@@ -354,7 +352,7 @@ __global__ void k_bn_bwd_apply(const ushort_t* __restrict__ x,
 inline int reduce_grid(int64_t M, int C) {
   const int TY = THREADS / (C >> 3);
   int64_t blocks = (M + TY - 1) / TY;
-  if (blocks > 1024) blocks = 1024;
+  if (blocks > 512) blocks = 512;
   if (blocks < 1) blocks = 1;
   return (int)blocks;
 }
@@ -370,22 +368,24 @@ inline int apply_grid(int64_t total8) {
 
 extern "C" {
 
-void bn_fwd_reduce(const ushort_t* x, float* shadows, int64_t M, int C,
+int bn_reduce_nblocks(int64_t M, int C) { return reduce_grid(M, C); }
+
+void bn_fwd_reduce(const ushort_t* x, float* partials, int64_t M, int C,
                    hipStream_t s) {
   hipLaunchKernelGGL(k_bn_fwd_reduce, dim3(reduce_grid(M, C)), dim3(THREADS),
-                     0, s, x, shadows, M, C);
+                     0, s, x, partials, M, C);
 }
 
-void bn_fwd_finalize(const float* shadows, const float* gamma,
+void bn_fwd_finalize(const float* partials, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
                      float* smean, float* sinvstd, float* scale_shift,
                      double momentum, double eps, int64_t M, int C,
                      bool update_running, hipStream_t s) {
   const int blocks = (C + THREADS - 1) / THREADS;
   hipLaunchKernelGGL(k_bn_fwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
-                     shadows, gamma, beta, rmean, rvar, smean, sinvstd,
+                     partials, gamma, beta, rmean, rvar, smean, sinvstd,
                      scale_shift, (float)momentum, (float)eps, M, C,
-                     update_running);
+                     reduce_grid(M, C), update_running);
 }
 
 void bn_eval_prep(const float* rmean, const float* rvar, const float* gamma,
@@ -410,25 +410,25 @@ void bn_fwd_apply(const ushort_t* x, const ushort_t* res, ushort_t* y,
 }
 
 void bn_bwd_reduce(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
-                   const float* smean, const float* sinvstd, float* shadows,
+                   const float* smean, const float* sinvstd, float* partials,
                    int64_t M, int C, bool relu, hipStream_t s) {
   const dim3 grid(reduce_grid(M, C));
   if (relu)
     hipLaunchKernelGGL((k_bn_bwd_reduce<true>), grid, dim3(THREADS), 0, s, x,
-                       dy, y, smean, sinvstd, shadows, M, C);
+                       dy, y, smean, sinvstd, partials, M, C);
   else
     hipLaunchKernelGGL((k_bn_bwd_reduce<false>), grid, dim3(THREADS), 0, s, x,
-                       dy, y, smean, sinvstd, shadows, M, C);
+                       dy, y, smean, sinvstd, partials, M, C);
 }
 
-void bn_bwd_finalize(const float* shadows, const float* gamma,
+void bn_bwd_finalize(const float* partials, const float* gamma,
                      const float* smean, const float* sinvstd, float* dgamma,
                      float* dbeta, float* coef, int64_t M, int C,
                      bool training, hipStream_t s) {
   const int blocks = (C + THREADS - 1) / THREADS;
   hipLaunchKernelGGL(k_bn_bwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
-                     shadows, gamma, smean, sinvstd, dgamma, dbeta, coef, M,
-                     C, training);
+                     partials, gamma, smean, sinvstd, dgamma, dbeta, coef, M,
+                     C, reduce_grid(M, C), training);
 }
 
 void bn_bwd_apply(const ushort_t* x, const ushort_t* dy, const ushort_t* y,
