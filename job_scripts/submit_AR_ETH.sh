@@ -1,0 +1,26 @@
+#!/bin/bash
+#SBATCH --job-name=AR_ETH
+#SBATCH --output=AR_ETH.out
+#SBATCH --error=AR_ETH.err
+#SBATCH --nodes=NB_NODES
+#SBATCH --cpus-per-task=32
+#SBATCH --gres=gpu:8
+#SBATCH --time=30:00:00
+#SBATCH --signal=B:USR1@120
+
+# Replace NB_NODES with the number of nodes to use.
+# One SLURM task per node; per-node batch 256 (reference recipe).
+# RCCL honors NCCL_* env; the trainer pins the fabric via
+# --network_interface_type.
+
+export HSA_ENABLE_IPC_MODE_LEGACY=0
+
+srun python -u gossip_sgd.py \
+    --batch_size 256 --lr 0.1 --num_dataloader_workers 16 \
+    --num_epochs 90 --nesterov True --warmup True \
+    --schedule 30 0.1 60 0.1 80 0.1 \
+    --train_fast False --master_port 40100 \
+    --tag 'AR_ETH' --print_freq 100 --verbose False \
+    --seed 1 --checkpoint_dir ./checkpoints/ \
+    --dataset imagefolder --dataset_dir $IMAGENET_DIR \
+    --network_interface_type 'ethernet' --all_reduce True --graph_type -1
