@@ -164,10 +164,14 @@ class GossipDataParallel(Module):
         self.gossip_ps_weight = self.ps_weight.clone()
 
         if not self.distributed:
-            # single-process job: nothing to gossip; wrapper is transparent
+            # nothing to gossip (single node-rank); the wrapper is
+            # transparent except for the intra-node multi-process tier,
+            # which still needs its param-broadcast / grad-reduce hooks
             self.lazy_mixing = True
             self.lazy_ps_factor = self.gossip_ps_factor.clone()
             self.gossip_thread = None
+            if self.nprocs_per_node > 1:
+                self.__register_hooks()
             return
 
         # staging + comm buffers (flat; reference used per-tensor clones,
@@ -274,7 +278,7 @@ class GossipDataParallel(Module):
         # copies in place, so the flat buffer is already current
 
     def forward(self, *inputs, **kwargs):
-        if self.distributed and self.nprocs_per_node > 1:
+        if self.nprocs_per_node > 1:
             self._sync_params_multiprocess()
         return self.module(*inputs, **kwargs)
 
