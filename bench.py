@@ -35,7 +35,7 @@ def parse_args():
     )
     p.add_argument("--peers-per-itr", type=int, default=1)
     p.add_argument(
-        "--norm", type=str, default="miopen",
+        "--norm", type=str, default="fused",
         help="batch-norm backend: miopen|native|fused",
     )
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
@@ -44,6 +44,11 @@ def parse_args():
     p.add_argument(
         "--no-graph", action="store_true",
         help="disable hipGraph capture of the local train step",
+    )
+    p.add_argument(
+        "--gossip-dtype", type=str, default="bf16",
+        choices=["bf16", "fp32"],
+        help="wire format of gossip messages (bf16 halves xGMI bytes)",
     )
     return p.parse_args()
 
@@ -108,6 +113,9 @@ def main():
             overlap=args.algorithm == "osgp",
             rank=rank if world_size > 1 else 0,
             world_size=world_size,
+            gossip_dtype=(
+                torch.bfloat16 if args.gossip_dtype == "bf16" else None
+            ),
         )
         model = gdp
         opt = FusedSGD(
@@ -225,6 +233,7 @@ def main():
                 "gossip_ms_per_step": round(gossip_ms, 3),
                 "channels_last": not args.no_channels_last,
                 "norm": args.norm,
+                "gossip_dtype": args.gossip_dtype,
             },
         }
         print(json.dumps(result), flush=True)

@@ -78,6 +78,7 @@ class GossipDataParallel(Module):
         nprocs_per_node: int = 1,
         local_node_group=None,
         flatten_grads: bool = True,
+        gossip_dtype: Optional[torch.dtype] = None,
     ):
         super().__init__()
 
@@ -174,11 +175,27 @@ class GossipDataParallel(Module):
                 self.__register_hooks()
             return
 
+        # wire format: gossip messages may travel narrower than the
+        # fp32 master params (bf16 halves xGMI bytes); the cast fuses
+        # into the pack/accumulate kernels.  Only regular mixing is
+        # allowed then (non-regular graphs transmit the push-sum weight
+        # in the message dtype, which must stay fp32).
+        self.gossip_dtype = gossip_dtype or first_param_dtype
+        if self.gossip_dtype != first_param_dtype:
+            assert self.dist_config["mixing"].is_regular(), (
+                "narrow gossip_dtype requires regular (uniform) mixing"
+            )
+
         # staging + comm buffers (flat; reference used per-tensor clones,
         # distributed.py:149-155)
-        self.gossip_device_buffer = torch.empty_like(self.flatp.flat)
+        self.gossip_device_buffer = torch.empty(
+            self.flatp.numel(), dtype=self.gossip_dtype,
+            device=self.flatp.flat.device,
+        )
         if self.__cpu_comm:
-            staged = self.flatp.flat.detach().cpu().clone()
+            staged = torch.empty(
+                self.flatp.numel(), dtype=self.gossip_dtype, device="cpu"
+            )
             if torch.cuda.is_available():
                 staged = staged.pin_memory()
             self.gossip_params = staged
@@ -388,13 +405,15 @@ class GossipDataParallel(Module):
             self.ps_weight += self.gossip_ps_weight
             if self.lazy_mixing:
                 self.ps_weight *= self.lazy_ps_factor
-                ops.add_scale_(
+                ops.add_scale_cast_(
                     self.flatp.flat,
                     self.gossip_device_buffer,
                     self.lazy_ps_factor.to(self.flatp.flat.dtype),
                 )
             else:
-                ops.add_scale_(self.flatp.flat, self.gossip_device_buffer, 1.0)
+                ops.add_scale_cast_(
+                    self.flatp.flat, self.gossip_device_buffer, 1.0
+                )
 
             self.logger.debug(f"updated ps-weight {self.ps_weight}")
             self.gossip_flag.clear()
@@ -427,13 +446,15 @@ class GossipDataParallel(Module):
         # in one pass (reference distributed.py:409-418 did ~161 mul_ +
         # ~161 copy_)
         if mix:
-            ops.pack_mix_(
+            ops.pack_mix_cast_(
                 self.flatp.flat,
                 self.gossip_device_buffer,
                 self.gossip_ps_factor.to(self.flatp.flat.dtype),
             )
         else:
-            ops.pack_mix_(self.flatp.flat, self.gossip_device_buffer, 1.0)
+            ops.pack_mix_cast_(
+                self.flatp.flat, self.gossip_device_buffer, 1.0
+            )
 
         if self._cuda:
             # hand the staging buffer to the gossip stream; async copy to

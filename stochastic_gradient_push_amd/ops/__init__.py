@@ -107,6 +107,40 @@ def pack_mix_(x: torch.Tensor, out: torch.Tensor, a: Scalar = 1.0) -> torch.Tens
     return out
 
 
+def pack_mix_cast_(
+    x: torch.Tensor, out: torch.Tensor, a: Scalar = 1.0
+) -> torch.Tensor:
+    """x *= a; out = x.to(out.dtype) — wire-format pack.  When ``out`` is
+    bf16 the cast fuses into the pack pass (half the xGMI bytes per
+    gossip message)."""
+    if out.dtype == x.dtype:
+        return pack_mix_(x, out, a)
+    assert out.dtype == torch.bfloat16 and x.dtype == torch.float32
+    if x.is_cuda:
+        _ext_for(x).pack_mix_bf16_(x, out, _as_scalar_tensor(a, x))
+    else:
+        if not (isinstance(a, float) and a == 1.0):
+            x.mul_(a if isinstance(a, float) else a.to(x.dtype))
+        out.copy_(x)
+    return out
+
+
+def add_scale_cast_(
+    x: torch.Tensor, r: torch.Tensor, a: Scalar = 1.0
+) -> torch.Tensor:
+    """x = (x + r.to(x.dtype)) * a — wire-format accumulate."""
+    if r.dtype == x.dtype:
+        return add_scale_(x, r, a)
+    assert r.dtype == torch.bfloat16 and x.dtype == torch.float32
+    if x.is_cuda:
+        _ext_for(x).add_scale_bf16_(x, r, _as_scalar_tensor(a, x))
+    else:
+        x.add_(r.to(x.dtype))
+        if not (isinstance(a, float) and a == 1.0):
+            x.mul_(a if isinstance(a, float) else a.to(x.dtype))
+    return x
+
+
 def average_(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """x = (x + y) * 0.5 (bilateral gossip merge)."""
     if x.is_cuda:

@@ -39,6 +39,10 @@ void sgp_add_scale(float* x, const float* r, const float* a, int64_t n,
 void sgp_pack_mix(float* x, float* out, const float* a, int64_t n,
                   hipStream_t stream);
 void sgp_average(float* x, const float* y, int64_t n, hipStream_t stream);
+void sgp_pack_mix_bf16(float* x, unsigned short* out, const float* a,
+                       int64_t n, hipStream_t stream);
+void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
+                        int64_t n, hipStream_t stream);
 void sgp_sgd_step(float* p, const float* g, float* buf, double lr, double mu,
                   double wd, double damp, bool nesterov, bool first, int64_t n,
                   hipStream_t stream);
@@ -107,6 +111,30 @@ void sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
   sgp_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
                lr, momentum, weight_decay, dampening, nesterov, first_step,
                p.numel(), current_stream(p));
+}
+
+void pack_mix_bf16_(torch::Tensor x, torch::Tensor out, torch::Tensor a) {
+  check_flat(x, "x");
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous()
+              && out.scalar_type() == torch::kBFloat16,
+              "out must be contiguous bf16 device tensor");
+  check_scalar(a, x);
+  TORCH_CHECK(x.numel() == out.numel(), "size mismatch");
+  sgp_pack_mix_bf16(x.data_ptr<float>(),
+                    reinterpret_cast<unsigned short*>(out.data_ptr()),
+                    a.data_ptr<float>(), x.numel(), current_stream(x));
+}
+
+void add_scale_bf16_(torch::Tensor x, torch::Tensor r, torch::Tensor a) {
+  check_flat(x, "x");
+  TORCH_CHECK(r.is_cuda() && r.is_contiguous()
+              && r.scalar_type() == torch::kBFloat16,
+              "r must be contiguous bf16 device tensor");
+  check_scalar(a, x);
+  TORCH_CHECK(x.numel() == r.numel(), "size mismatch");
+  sgp_add_scale_bf16(x.data_ptr<float>(),
+                     reinterpret_cast<const unsigned short*>(r.data_ptr()),
+                     a.data_ptr<float>(), x.numel(), current_stream(x));
 }
 
 // ---------------------------------------------------------------- BN ops
@@ -251,6 +279,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_mix_", &pack_mix_, "x *= a; out = x");
   m.def("average_", &average_, "x = (x + y) / 2");
   m.def("sgd_step_", &sgd_step_, "fused momentum-SGD step");
+  m.def("pack_mix_bf16_", &pack_mix_bf16_,
+        "x *= a; out_bf16 = bf16(x) (wire-format pack)");
+  m.def("add_scale_bf16_", &add_scale_bf16_,
+        "x = (x + float(r_bf16)) * a (wire-format accumulate)");
   m.def("bn_partials_numel", &bn_partials_numel);
   m.def("bn_fwd_reduce", &bn_fwd_reduce_py);
   m.def("bn_fwd_finalize", &bn_fwd_finalize_py);

@@ -108,6 +108,70 @@ __global__ void k_average(float* __restrict__ x, const float* __restrict__ y,
     x[j] = (x[j] + y[j]) * 0.5f;
 }
 
+// -------------------------------------------- bf16 wire-format variants
+// Gossip messages can travel as bf16 (halves xGMI bytes per exchange);
+// master params stay fp32.  Conversion is RNE, fused into the pack /
+// accumulate pass.
+
+__device__ __forceinline__ unsigned short f2b_(float f) {
+  union { unsigned int i; float f; } v;
+  v.f = f;
+  unsigned int r = v.i + 0x7FFFu + ((v.i >> 16) & 1u);
+  return (unsigned short)(r >> 16);
+}
+
+__device__ __forceinline__ float b2f_(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+
+// t = x * *a ; x = t ; out_bf16 = bf16(t)
+__global__ void k_pack_mix_bf16(float* __restrict__ x,
+                                unsigned short* __restrict__ out,
+                                const float* __restrict__ a,
+                                int64_t n4, int64_t n) {
+  const float s = *a;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n4; i += stride) {
+    float4 v = ld4(x + 4 * i);
+    v.x *= s; v.y *= s; v.z *= s; v.w *= s;
+    st4(x + 4 * i, v);
+    ushort4 o;
+    o.x = f2b_(v.x); o.y = f2b_(v.y); o.z = f2b_(v.z); o.w = f2b_(v.w);
+    *reinterpret_cast<ushort4*>(out + 4 * i) = o;
+  }
+  for (int64_t j = 4 * n4 + (blockIdx.x * blockDim.x + threadIdx.x);
+       j < n; j += stride) {
+    const float t = x[j] * s;
+    x[j] = t;
+    out[j] = f2b_(t);
+  }
+}
+
+// x = (x + float(r_bf16)) * *a
+__global__ void k_add_scale_bf16(float* __restrict__ x,
+                                 const unsigned short* __restrict__ r,
+                                 const float* __restrict__ a,
+                                 int64_t n4, int64_t n) {
+  const float s = *a;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n4; i += stride) {
+    float4 v = ld4(x + 4 * i);
+    ushort4 u = *reinterpret_cast<const ushort4*>(r + 4 * i);
+    v.x = (v.x + b2f_(u.x)) * s;
+    v.y = (v.y + b2f_(u.y)) * s;
+    v.z = (v.z + b2f_(u.z)) * s;
+    v.w = (v.w + b2f_(u.w)) * s;
+    st4(x + 4 * i, v);
+  }
+  for (int64_t j = 4 * n4 + (blockIdx.x * blockDim.x + threadIdx.x);
+       j < n; j += stride)
+    x[j] = (x[j] + b2f_(r[j])) * s;
+}
+
 // ------------------------------------------------------------- sgd_step_
 // torch.optim.SGD semantics over flat buffers, one pass:
 //   d   = g + wd * p
@@ -197,6 +261,20 @@ void sgp_average(float* x, const float* y, int64_t n, hipStream_t stream) {
   const int64_t n4 = n / 4;
   hipLaunchKernelGGL(k_average, dim3(grid_for(n4 ? n4 : n)), dim3(THREADS), 0,
                      stream, x, y, n4, n);
+}
+
+void sgp_pack_mix_bf16(float* x, unsigned short* out, const float* a,
+                       int64_t n, hipStream_t stream) {
+  const int64_t n4 = n / 4;
+  hipLaunchKernelGGL(k_pack_mix_bf16, dim3(grid_for(n4 ? n4 : n)),
+                     dim3(THREADS), 0, stream, x, out, a, n4, n);
+}
+
+void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
+                        int64_t n, hipStream_t stream) {
+  const int64_t n4 = n / 4;
+  hipLaunchKernelGGL(k_add_scale_bf16, dim3(grid_for(n4 ? n4 : n)),
+                     dim3(THREADS), 0, stream, x, r, a, n4, n);
 }
 
 void sgp_sgd_step(float* p, const float* g, float* buf, double lr, double mu,

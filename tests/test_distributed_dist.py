@@ -203,3 +203,38 @@ def _overlap_sgp(rank, world_size):
 
 def test_overlap_sgp_consensus():
     run_dist(_overlap_sgp, world_size=2)
+
+
+def _bf16_wire_consensus(rank, world_size):
+    """bf16 wire-format gossip still reaches consensus (zero lr)."""
+    from stochastic_gradient_push_amd import GossipDataParallel
+
+    model = tiny_model(seed=rank)
+    flat0 = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    target = flat0.clone()
+    dist.all_reduce(target)
+    target /= world_size
+
+    gdp = GossipDataParallel(
+        model, push_sum=True, gossip_dtype=torch.bfloat16,
+    )
+    opt = torch.optim.SGD(gdp.parameters(), lr=0.0)
+    loss_fn = nn.CrossEntropyLoss()
+    gdp.train()
+    x = torch.randn(2, 3, 8, 8)
+    y = torch.randint(0, 10, (2,))
+    for _ in range(30):
+        _train_step(gdp, opt, x, y, loss_fn)
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    gdp.unbias()
+    flat = gdp.flatp.flat.detach()
+    # bf16 wire: consensus up to bf16 resolution
+    assert torch.allclose(flat, target, atol=3e-2), (
+        f"rank {rank}: max err {(flat - target).abs().max()}"
+    )
+    gdp.shutdown()
+
+
+def test_bf16_wire_consensus():
+    run_dist(_bf16_wire_consensus, world_size=2)

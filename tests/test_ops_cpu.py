@@ -73,3 +73,18 @@ def test_sgd_step_matches_torch(momentum, nesterov):
             weight_decay=1e-4, nesterov=nesterov, first_step=(step == 0),
         )
         assert torch.allclose(p, p_ref.detach(), atol=1e-6), f"step {step}"
+
+
+def test_pack_mix_cast_and_add_scale_cast():
+    x = torch.randn(65)
+    out = torch.empty(65, dtype=torch.bfloat16)
+    ref = x * 0.5
+    ops.pack_mix_cast_(x, out, 0.5)
+    assert torch.allclose(x, ref)
+    assert torch.allclose(out.float(), ref, atol=1e-2, rtol=1e-2)
+
+    y = torch.randn(65)
+    r = torch.randn(65).to(torch.bfloat16)
+    ref2 = (y + r.float()) * 0.25
+    ops.add_scale_cast_(y, r, 0.25)
+    assert torch.allclose(y, ref2, atol=1e-6)

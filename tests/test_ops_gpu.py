@@ -138,3 +138,24 @@ def test_fused_sgd_matches_torch_on_model():
     torch.cuda.synchronize()
     for p, q in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p, q, atol=1e-5)
+
+
+@pytest.mark.parametrize("n", [255, (1 << 20) + 3])
+def test_pack_mix_bf16_gpu(n):
+    x = torch.randn(n, device=dev())
+    out = torch.empty(n, dtype=torch.bfloat16, device=dev())
+    ref = x * 0.5
+    ops.pack_mix_cast_(x, out, torch.tensor([0.5], device=dev()))
+    torch.cuda.synchronize()
+    assert torch.allclose(x, ref)
+    assert torch.allclose(out.float(), ref.to(torch.bfloat16).float())
+
+
+@pytest.mark.parametrize("n", [255, (1 << 20) + 3])
+def test_add_scale_bf16_gpu(n):
+    x = torch.randn(n, device=dev())
+    r = torch.randn(n, device=dev()).to(torch.bfloat16)
+    ref = (x + r.float()) * 0.25
+    ops.add_scale_cast_(x, r, torch.tensor([0.25], device=dev()))
+    torch.cuda.synchronize()
+    assert torch.allclose(x, ref, atol=1e-6)
