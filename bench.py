@@ -46,6 +46,12 @@ def parse_args():
         help="disable hipGraph capture of the local train step",
     )
     p.add_argument(
+        "--opt", type=str, default="fused",
+        choices=["fused", "foreach"],
+        help="fused flat-buffer SGD vs torch foreach SGD with steal-mode "
+             "grads (no pre-wired flat grads, no accumulate adds)",
+    )
+    p.add_argument(
         "--gossip-dtype", type=str, default="bf16",
         choices=["bf16", "fp32"],
         help="wire format of gossip messages (bf16 halves xGMI bytes)",
@@ -116,11 +122,18 @@ def main():
             gossip_dtype=(
                 torch.bfloat16 if args.gossip_dtype == "bf16" else None
             ),
+            flatten_grads=(args.opt == "fused"),
         )
         model = gdp
-        opt = FusedSGD(
-            gdp.flatp, lr=0.1, momentum=0.9, weight_decay=1e-4,
-        )
+        if args.opt == "fused":
+            opt = FusedSGD(
+                gdp.flatp, lr=0.1, momentum=0.9, weight_decay=1e-4,
+            )
+        else:
+            opt = torch.optim.SGD(
+                model.parameters(), lr=0.1, momentum=0.9,
+                weight_decay=1e-4, foreach=True,
+            )
 
     loss_fn = nn.CrossEntropyLoss()
     model.train()

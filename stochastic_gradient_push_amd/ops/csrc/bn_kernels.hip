@@ -139,12 +139,12 @@ __global__ void k_bn_fwd_finalize(const float* __restrict__ partials,
                                   float* __restrict__ scale_shift,
                                   float momentum, float eps, int64_t M,
                                   int C, int nblocks, bool update_running) {
-  const int cl = threadIdx.x % 32;
-  const int w = threadIdx.x / 32;  // 8 slices
-  const int c = blockIdx.x * 32 + cl;
+  const int cl = threadIdx.x % 8;
+  const int w = threadIdx.x / 8;  // 32 slices
+  const int c = blockIdx.x * 8 + cl;
   float s = 0.f, q = 0.f;
   if (c < C) {
-    for (int k = w; k < nblocks; k += 8) {
+    for (int k = w; k < nblocks; k += 32) {
       s += partials[(int64_t)k * 2 * C + c];
       q += partials[(int64_t)k * 2 * C + C + c];
     }
@@ -154,9 +154,9 @@ __global__ void k_bn_fwd_finalize(const float* __restrict__ partials,
   red[256 + threadIdx.x] = q;
   __syncthreads();
   if (w != 0 || c >= C) return;
-  for (int t = 1; t < 8; ++t) {
-    s += red[t * 32 + cl];
-    q += red[256 + t * 32 + cl];
+  for (int t = 1; t < 32; ++t) {
+    s += red[t * 8 + cl];
+    q += red[256 + t * 8 + cl];
   }
   const float inv_m = 1.0f / (float)M;
   const float mean = s * inv_m;
@@ -305,25 +305,24 @@ __global__ void k_bn_bwd_finalize(const float* __restrict__ partials,
                                   float* __restrict__ dbeta,
                                   float* __restrict__ coef, int64_t M,
                                   int C, int nblocks, bool training) {
-  const int c = blockIdx.x * 32 + threadIdx.x % 32;
-  const int cl = threadIdx.x % 32;
-  const int w = threadIdx.x / 32;
-  const int c2 = blockIdx.x * 32 + cl;
+  const int cl = threadIdx.x % 8;
+  const int w = threadIdx.x / 8;  // 32 slices
+  const int c = blockIdx.x * 8 + cl;
   float dg = 0.f, db = 0.f;
-  if (c2 < C) {
-    for (int k = w; k < nblocks; k += 8) {
-      dg += partials[(int64_t)k * 2 * C + c2];
-      db += partials[(int64_t)k * 2 * C + C + c2];
+  if (c < C) {
+    for (int k = w; k < nblocks; k += 32) {
+      dg += partials[(int64_t)k * 2 * C + c];
+      db += partials[(int64_t)k * 2 * C + C + c];
     }
   }
   __shared__ float red[256 * 2];
   red[threadIdx.x] = dg;
   red[256 + threadIdx.x] = db;
   __syncthreads();
-  if (w != 0 || c2 >= C) return;
-  for (int t = 1; t < 8; ++t) {
-    dg += red[t * 32 + cl];
-    db += red[256 + t * 32 + cl];
+  if (w != 0 || c >= C) return;
+  for (int t = 1; t < 32; ++t) {
+    dg += red[t * 8 + cl];
+    db += red[256 + t * 8 + cl];
   }
   dgamma[c] = dg;
   dbeta[c] = db;
@@ -410,7 +409,7 @@ void bn_fwd_finalize(const float* partials, const float* gamma,
                      float* smean, float* sinvstd, float* scale_shift,
                      double momentum, double eps, int64_t M, int C,
                      bool update_running, hipStream_t s) {
-  const int blocks = (C + 31) / 32;
+  const int blocks = (C + 7) / 8;
   hipLaunchKernelGGL(k_bn_fwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
                      partials, gamma, beta, rmean, rvar, smean, sinvstd,
                      scale_shift, (float)momentum, (float)eps, M, C,
@@ -454,7 +453,7 @@ void bn_bwd_finalize(const float* partials, const float* gamma,
                      const float* smean, const float* sinvstd, float* dgamma,
                      float* dbeta, float* coef, int64_t M, int C,
                      bool training, hipStream_t s) {
-  const int blocks = (C + 31) / 32;
+  const int blocks = (C + 7) / 8;
   hipLaunchKernelGGL(k_bn_bwd_finalize, dim3(blocks), dim3(THREADS), 0, s,
                      partials, gamma, smean, sinvstd, dgamma, dbeta, coef, M,
                      C, reduce_grid(M, C), training);
