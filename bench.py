@@ -173,7 +173,9 @@ def main():
             compute_step(inner)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        # thread_local: the (idle) gossip thread and NCCL watchdog must
+        # not poison a global-mode stream capture at world_size > 1
+        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
             compute_step(inner)
 
         def step():
