@@ -31,7 +31,9 @@ def parse_args():
     p.add_argument("--model", type=str, default="resnet50")
     p.add_argument(
         "--algorithm", type=str, default="sgp",
-        choices=["sgp", "osgp", "dpsgd", "ar", "adpsgd"],
+        choices=["sgp", "osgp", "dpsgd", "ar"],
+        help="AD-PSGD throughput is measured via gossip_sgd_adpsgd.py "
+             "(its trainer process does not join this dist world)",
     )
     p.add_argument("--peers-per-itr", type=int, default=1)
     p.add_argument(

@@ -23,9 +23,11 @@ ext = CUDAExtension(
     name="stochastic_gradient_push_amd.ops._gossip_kernels",
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "comm_core.cpp"),
         os.path.join(CSRC, "gossip_kernels.hip"),
         os.path.join(CSRC, "bn_kernels.hip"),
     ],
+    libraries=["rccl"],
     extra_compile_args={
         "cxx": ["-O3"],
         "nvcc": ["-O3", "--offload-arch=gfx950"],

@@ -79,6 +79,7 @@ class GossipDataParallel(Module):
         local_node_group=None,
         flatten_grads: bool = True,
         gossip_dtype: Optional[torch.dtype] = None,
+        comm_backend: str = "c10d",
     ):
         super().__init__()
 
@@ -208,6 +209,16 @@ class GossipDataParallel(Module):
         self.gossip_group = create_process_group(
             list(range(dist.get_world_size()))
         )
+
+        # optional native comm core: its own RCCL communicator + HIP
+        # stream, built collectively HERE (main thread) and driven by the
+        # gossip thread
+        assert comm_backend in ("c10d", "rccl")
+        self.dist_config["transport"] = None
+        if comm_backend == "rccl":
+            from .comm import create_rccl_transport
+
+            self.dist_config["transport"] = create_rccl_transport()
 
         # control objects (reference distributed.py:157-165)
         self.gossip_lock = threading.Lock()
@@ -518,6 +529,7 @@ class GossipDataParallel(Module):
             world_size=dist_config["world_size"],
             logger=logger,
             group=gossip_group,
+            transport=dist_config.get("transport"),
         )
         dist_config["gossipers"] = {gossip_params.dtype: gossiper}
         gossip_ps_factor.data.copy_(gossiper.mixing_weights["lo"])

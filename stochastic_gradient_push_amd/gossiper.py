@@ -76,6 +76,7 @@ class Gossiper:
         rank: Optional[int] = None,
         world_size: Optional[int] = None,
         group=None,
+        transport=None,
     ):
         """
         :param msg: prototype message tensor (sized like the flat params)
@@ -88,6 +89,10 @@ class Gossiper:
             concurrently with main-thread collectives on a separate RCCL
             communicator (the reference achieved this implicitly via its
             per-edge groups).
+        :param transport: optional native
+            :class:`~stochastic_gradient_push_amd.comm.RcclTransport`;
+            when given, exchanges bypass c10d and go straight through the
+            C++ comm core's own RCCL communicator.
         """
         self.logger = logger
         if rank is None or world_size is None:
@@ -97,6 +102,7 @@ class Gossiper:
         self.rank = rank
         self.world_size = world_size
         self.group = group
+        self.transport = transport
 
         assert isinstance(graph, GraphManager)
         self._graph_manager = graph
@@ -270,16 +276,30 @@ class Gossiper:
                 self.in_msg_buffer.zero_()
             recvs = self._recv_buffers(len(self.in_edges))
 
-        ops = []
-        for e in self.out_edges:
-            assert e.src == self.rank
-            ops.append(dist.P2POp(dist.isend, send_msg, e.dest, group=self.group))
-        for buf, e in zip(recvs, self.in_edges):
-            ops.append(dist.P2POp(dist.irecv, buf, e.src, group=self.group))
-        if ops:
-            reqs = dist.batch_isend_irecv(ops)
-            for r in reqs:
-                r.wait()
+        if self.transport is not None:
+            for e in self.out_edges:
+                assert e.src == self.rank
+            self.transport.exchange(
+                send_msg,
+                [e.dest for e in self.out_edges],
+                recvs,
+                [e.src for e in self.in_edges],
+            )
+        else:
+            ops = []
+            for e in self.out_edges:
+                assert e.src == self.rank
+                ops.append(
+                    dist.P2POp(dist.isend, send_msg, e.dest, group=self.group)
+                )
+            for buf, e in zip(recvs, self.in_edges):
+                ops.append(
+                    dist.P2POp(dist.irecv, buf, e.src, group=self.group)
+                )
+            if ops:
+                reqs = dist.batch_isend_irecv(ops)
+                for r in reqs:
+                    r.wait()
         if not direct:
             for buf in recvs:
                 self.in_msg_buffer.add_(buf)

@@ -272,7 +272,10 @@ void bn_bwd_apply_py(torch::Tensor x, torch::Tensor dy,
 
 }  // namespace
 
+void init_comm_core(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  init_comm_core(m);
   m.doc() = "gfx950 fused gossip + batchnorm kernels";
   m.def("scale_", &scale_, "x *= a (in place, fused over flat buffer)");
   m.def("add_scale_", &add_scale_, "x = (x + r) * a");

@@ -159,3 +159,21 @@ def test_add_scale_bf16_gpu(n):
     ops.add_scale_cast_(x, r, torch.tensor([0.25], device=dev()))
     torch.cuda.synchronize()
     assert torch.allclose(x, ref, atol=1e-6)
+
+
+def test_rccl_transport_self_exchange():
+    """Native comm core: 1-rank communicator, self send/recv round-trip
+    (RCCL supports self-p2p inside a group as a device copy)."""
+    from stochastic_gradient_push_amd.comm import create_rccl_transport
+    from stochastic_gradient_push_amd.ops import _gossip_kernels as k
+
+    uid = k.rccl_unique_id()
+    t = create_rccl_transport(
+        device_index=0, rank=0, world_size=1, unique_id=uid
+    )
+    for dtype in (torch.float32, torch.bfloat16):
+        x = torch.randn(1 << 16, device=dev()).to(dtype)
+        r = torch.zeros_like(x)
+        t.exchange(x, [0], [r], [0])
+        torch.cuda.synchronize()
+        assert torch.equal(x, r)
