@@ -113,6 +113,33 @@ class GossipDataParallel(Module):
         first_param_dtype = next(module.parameters()).dtype
         self._cuda = next(module.parameters()).is_cuda
 
+        # single-process multi-GPU replica tier (reference
+        # distributed.py:91-99, 231-254, 524-549): DataParallel-style
+        # replicas driven by one process.  The MI355X-native mode is one
+        # process per GPU; this tier exists for API completeness.
+        self.device_ids = device_ids
+        self.output_device = device_ids[0] if device_ids else None
+        self._replica_tier = bool(device_ids) and len(device_ids) > 1
+        if self._replica_tier:
+            import copy as _copy
+
+            assert self._cuda, "replica tier requires CUDA modules"
+            self._module_copies = [module]
+            for d in device_ids[1:]:
+                rep = _copy.deepcopy(module).to(torch.device("cuda", d))
+                for p, rp in zip(module.parameters(), rep.parameters()):
+                    rp.requires_grad = p.requires_grad
+                self._module_copies.append(rep)
+            from .ops.flat import FlatParams as _FP
+
+            self._replica_flat = [
+                _FP(rep, flatten_grads=True)
+                for rep in self._module_copies[1:]
+            ]
+        else:
+            self._module_copies = [module]
+            self._replica_flat = []
+
         # communication device (reference distributed.py:101-105)
         if comm_device is None:
             if dist.is_initialized():
@@ -172,7 +199,7 @@ class GossipDataParallel(Module):
             self.lazy_mixing = True
             self.lazy_ps_factor = self.gossip_ps_factor.clone()
             self.gossip_thread = None
-            if self.nprocs_per_node > 1:
+            if self.nprocs_per_node > 1 or self._replica_tier:
                 self.__register_hooks()
             return
 
@@ -308,15 +335,59 @@ class GossipDataParallel(Module):
     def forward(self, *inputs, **kwargs):
         if self.nprocs_per_node > 1:
             self._sync_params_multiprocess()
+        if self._replica_tier:
+            from torch.nn.parallel.parallel_apply import parallel_apply
+            from torch.nn.parallel.scatter_gather import (
+                gather,
+                scatter_kwargs,
+            )
+
+            inputs, kwargs = scatter_kwargs(
+                inputs, kwargs, self.device_ids, dim=0
+            )
+            self._sync_replica_params()
+            outputs = parallel_apply(
+                self._module_copies[: len(inputs)], inputs, kwargs,
+                self.device_ids[: len(inputs)],
+            )
+            return gather(outputs, self.output_device, dim=0)
         return self.module(*inputs, **kwargs)
+
+    def _sync_replica_params(self):
+        """Push master params+buffers to every replica (reference
+        distributed.py:256-276 used broadcast_coalesced; with flat
+        buffers each replica is ONE cross-device copy)."""
+        for rep_flat in self._replica_flat:
+            rep_flat.flat.copy_(self.flatp.flat, non_blocking=True)
+        master_buffers = list(self.module.buffers())
+        if master_buffers:
+            for rep in self._module_copies[1:]:
+                for mb, rb in zip(master_buffers, rep.buffers()):
+                    rb.copy_(mb, non_blocking=True)
+
+    def _reduce_replica_grads(self):
+        """Sum replica grads into the master's flat grad (reference
+        distributed.py:528-549 used reduce_add_coalesced)."""
+        for rep_flat in self._replica_flat:
+            if rep_flat.flat_grad is not None:
+                self.flatp.flat_grad.add_(
+                    rep_flat.flat_grad.to(
+                        self.flatp.flat_grad.device, non_blocking=True
+                    )
+                )
+                rep_flat.zero_grad()
 
     def train(self, mode: bool = True):
         super().train(mode)
+        for rep in self._module_copies[1:]:
+            rep.train(mode)
         self.gossip_enable = self.distributed and mode
         return self
 
     def eval(self):
         super().eval()
+        for rep in self._module_copies[1:]:
+            rep.eval()
         self.gossip_enable = False
         if self.distributed:
             self._query_gossip_queue(non_blocking=self.asynch)
@@ -601,6 +672,8 @@ class GossipDataParallel(Module):
 
     def __make_backward_hook(self):
         def hook(*unused):
+            if self._replica_tier:
+                self._reduce_replica_grads()
             if self.nprocs_per_node > 1 and self.local_node_group is not None:
                 # intra-node grad averaging over the local group; one
                 # all-reduce when grads are flat (reference
