@@ -149,3 +149,13 @@ def _bilat(rank, world_size):
 
 def test_bilat_pushpull_converges():
     run_dist(_bilat, world_size=2)
+
+
+@pytest.mark.parametrize("graph_cls", [
+    "NPeerDynamicDirectedExponentialGraph",
+    "DynamicBipartiteExponentialGraph",
+])
+def test_pushsum_eight_process_emulation(graph_cls):
+    """Multi-node emulation (SURVEY tier): 8 CPU processes, larger
+    topology, distributed averaging converges."""
+    run_dist(_pushsum_average, world_size=8, args=("PushSum", graph_cls))
