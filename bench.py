@@ -48,6 +48,11 @@ def parse_args():
         help="disable hipGraph capture of the local train step",
     )
     p.add_argument(
+        "--conv-impl", type=str, default="miopen",
+        choices=["miopen", "gemm"],
+        help="1x1 convs: MIOpen conv kernels vs hipBLASLt GEMM dispatch",
+    )
+    p.add_argument(
         "--opt", type=str, default="fused",
         choices=["fused", "foreach"],
         help="fused flat-buffer SGD vs torch foreach SGD with steal-mode "
@@ -93,7 +98,9 @@ def main():
     from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
 
     torch.manual_seed(1234 + rank)
-    model = build_resnet(args.model, norm=args.norm).to(device)
+    model = build_resnet(
+        args.model, norm=args.norm, conv_impl=args.conv_impl
+    ).to(device)
     if cuda and not args.no_channels_last:
         model = model.to(memory_format=torch.channels_last)
 
@@ -251,6 +258,7 @@ def main():
                 "channels_last": not args.no_channels_last,
                 "norm": args.norm,
                 "gossip_dtype": args.gossip_dtype,
+                "conv_impl": args.conv_impl,
             },
         }
         print(json.dumps(result), flush=True)

@@ -49,6 +49,39 @@ def make_norm(kind: str):
     raise ValueError(f"unknown norm kind {kind}")
 
 
+class GemmConv1x1(nn.Conv2d):
+    """1x1 convolution dispatched as a plain GEMM.
+
+    In channels_last a 1x1 conv IS ``y[M, Co] = x[M, Ci] @ W[Co, Ci]^T``
+    with M = N*H*W and zero-copy reshapes; ``torch.matmul`` routes it to
+    hipBLASLt's tuned bf16 MFMA GEMMs instead of MIOpen's conv kernels
+    (profiled far below the MFMA roofline at these shapes).  Stride-2
+    variants subsample rows first.  Autograd covers dgrad/wgrad as the
+    transposed GEMMs through the same library path.
+    """
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__(cin, cout, 1, stride=stride, bias=False)
+
+    def forward(self, x):
+        if not (
+            x.is_cuda
+            and x.dtype in (torch.bfloat16, torch.float16)
+            and x.is_contiguous(memory_format=torch.channels_last)
+        ):
+            return super().forward(x)
+        if self.stride[0] != 1:
+            x = x[:, :, :: self.stride[0], :: self.stride[1]]
+            x = x.contiguous(memory_format=torch.channels_last)
+        n, c, h, w = x.shape
+        x2d = x.permute(0, 2, 3, 1).reshape(n * h * w, c)  # zero-copy
+        weight = self.weight.view(self.out_channels, c).to(x.dtype)
+        y2d = torch.matmul(x2d, weight.t())
+        return (
+            y2d.view(n, h, w, self.out_channels).permute(0, 3, 1, 2)
+        )  # channels_last strides, zero-copy
+
+
 def _fused_supported(x: torch.Tensor, C: int) -> bool:
     """The hand-written kernels need bf16 NHWC with C = 8 * 2^k <= 2048
     (the reduce kernel's thread geometry; every ResNet width qualifies)."""

@@ -22,7 +22,12 @@ def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
-def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+def conv1x1(cin: int, cout: int, stride: int = 1,
+            impl: str = "miopen") -> nn.Conv2d:
+    if impl == "gemm":
+        from .layers import GemmConv1x1
+
+        return GemmConv1x1(cin, cout, stride=stride)
     return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
 
 
@@ -30,7 +35,8 @@ class BasicBlock(nn.Module):
     expansion = 1
 
     def __init__(self, cin, planes, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d, fused=False):
+                 norm_layer=nn.BatchNorm2d, fused=False,
+                 conv_impl="miopen"):
         super().__init__()
         self._fused = fused
         self.conv1 = conv3x3(cin, planes, stride)
@@ -66,12 +72,14 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, cin, planes, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d, fused=False):
+                 norm_layer=nn.BatchNorm2d, fused=False,
+                 conv_impl="miopen"):
         super().__init__()
         self._fused = fused
-        self.conv1 = conv1x1(cin, planes)
+        self.conv1 = conv1x1(cin, planes, impl=conv_impl)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        self.conv3 = conv1x1(planes, planes * self.expansion,
+                             impl=conv_impl)
         self.downsample = downsample
         self.stride = stride
         if fused:
@@ -111,11 +119,13 @@ class ResNet(nn.Module):
         num_classes: int = 1000,
         zero_init_residual: bool = True,
         norm: str = "miopen",
+        conv_impl: str = "miopen",
     ):
         super().__init__()
         from .layers import FusedBatchNorm2d, make_norm
 
         self._fused = norm == "fused"
+        self._conv_impl = conv_impl
         self._norm_layer = (
             (lambda c: FusedBatchNorm2d(c, relu=False))
             if self._fused else make_norm(norm)
@@ -159,16 +169,19 @@ class ResNet(nn.Module):
         downsample = None
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = nn.Sequential(
-                conv1x1(self.inplanes, planes * block.expansion, stride),
+                conv1x1(self.inplanes, planes * block.expansion, stride,
+                        impl=self._conv_impl),
                 self._norm_layer(planes * block.expansion),
             )
         layers = [block(self.inplanes, planes, stride, downsample,
-                        norm_layer=self._norm_layer, fused=self._fused)]
+                        norm_layer=self._norm_layer, fused=self._fused,
+                        conv_impl=self._conv_impl)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
             layers.append(block(self.inplanes, planes,
                                 norm_layer=self._norm_layer,
-                                fused=self._fused))
+                                fused=self._fused,
+                                conv_impl=self._conv_impl))
         return nn.Sequential(*layers)
 
     def forward(self, x):
@@ -193,10 +206,11 @@ _CONFIGS = {
 
 def build_resnet(
     name: str, num_classes: int = 1000, zero_init_residual: bool = True,
-    norm: str = "miopen",
+    norm: str = "miopen", conv_impl: str = "miopen",
 ) -> ResNet:
     block, layers = _CONFIGS[name]
-    return ResNet(block, layers, num_classes, zero_init_residual, norm=norm)
+    return ResNet(block, layers, num_classes, zero_init_residual,
+                  norm=norm, conv_impl=conv_impl)
 
 
 def resnet18(**kw) -> ResNet:
