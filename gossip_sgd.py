@@ -100,6 +100,9 @@ def build_parser():
     p.add_argument("--num_itr_ignore", type=int, default=10)
     p.add_argument("--dataset_dir", type=str, default=None)
     p.add_argument("--no_cuda_streams", action="store_true")
+    p.add_argument("--hip_graph", default="True", type=str,
+                   help="hipGraph-capture the local train step (cuda, "
+                        "fused_sgd, lazy-mixing algorithms)")
     # MI355X-native extensions
     p.add_argument("--dataset", default="synthetic",
                    choices=["synthetic", "imagefolder"])
@@ -144,7 +147,8 @@ def parse_args(argv=None):
 
     for flag in ("resume", "verbose", "train_fast", "nesterov",
                  "checkpoint_all", "warmup", "overlap", "push_sum",
-                 "all_reduce", "overwrite_checkpoints", "fused_sgd"):
+                 "all_reduce", "overwrite_checkpoints", "fused_sgd",
+                 "hip_graph"):
         setattr(args, flag, str2bool(getattr(args, flag)))
 
     args.rank, args.world_size = discover_rank_world(args)
@@ -351,9 +355,71 @@ def one_hot(target, num_classes, device):
     ).scatter_(1, target.view(-1, 1), 1)
 
 
+class GraphedStep:
+    """hipGraph-captured fwd+bwd+FusedSGD step for the trainer.
+
+    The gossip state machine (query/transfer, host flag logic, fused
+    residual merge) runs eagerly around the replay; push-sum scalars and
+    the learning rate are device-resident so replay sees live values.
+    Requires lazy mixing (sync SGP/D-PSGD) — bias/de-bias are no-ops
+    there, keeping the captured region pure compute.
+    """
+
+    WARMUP_STEPS = 3
+
+    def __init__(self, args, gdp, optimizer, criterion):
+        self.args = args
+        self.gdp = gdp
+        self.inner = gdp.module
+        self.optimizer = optimizer
+        self.criterion = criterion
+        self.graph = None
+        self.calls = 0
+        self.sx = None
+        self.st = None
+        self.out = None
+        self.loss = None
+
+    def _compute(self):
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            out = self.inner(self.sx)
+        loss = self.criterion(out, self.st)
+        loss.backward()
+        self.optimizer.step()
+        self.optimizer.zero_grad()
+        return out, loss
+
+    def __call__(self, batch, kl_target):
+        if self.sx is None:
+            self.sx = batch.clone()
+            self.st = kl_target.clone()
+        else:
+            self.sx.copy_(batch, non_blocking=True)
+            self.st.copy_(kl_target, non_blocking=True)
+        if self.graph is None:
+            if self.calls < self.WARMUP_STEPS:
+                self.calls += 1
+                out, loss = self._compute()
+            else:
+                torch.cuda.synchronize()
+                self.graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(
+                    self.graph, capture_error_mode="thread_local"
+                ):
+                    out, loss = self._compute()
+                self.out, self.loss = out, loss
+        else:
+            self.graph.replay()
+            out, loss = self.out, self.loss
+        if self.gdp.distributed:
+            self.gdp._query_gossip_queue(non_blocking=self.gdp.asynch)
+            self.gdp.transfer_params()
+        return out, loss
+
+
 def train_epoch(args, log, model, criterion, optimizer, batch_meter,
                 data_meter, nn_meter, loader, epoch, start_itr, begin_time,
-                num_itr_ignore):
+                num_itr_ignore, graphed_step=None):
     losses = Meter(ptag="Loss")
     top1 = Meter(ptag="Prec@1")
     top5 = Meter(ptag="Prec@5")
@@ -383,20 +449,27 @@ def train_epoch(args, log, model, criterion, optimizer, batch_meter,
             data_meter.update(time.time() - batch_time)
 
         nn_time = time.time()
-        with torch.autocast(
-            device_type=args.device, dtype=torch.bfloat16,
-            enabled=(args.device == "cuda"),
-        ):
-            output = model(batch)
-        loss = criterion(output, kl_target)
-        loss.backward()
-        if i % 100 == 0:
-            update_learning_rate(args, optimizer, epoch, itr=i,
-                                 itr_per_epoch=len(loader))
-        optimizer.step()
-        optimizer.zero_grad()
-        if is_gossip and not args.overlap:
-            model.transfer_params()
+        if graphed_step is not None:
+            if i % 100 == 0:
+                update_learning_rate(args, optimizer, epoch, itr=i,
+                                     itr_per_epoch=len(loader))
+                optimizer.sync_lr()
+            output, loss = graphed_step(batch, kl_target)
+        else:
+            with torch.autocast(
+                device_type=args.device, dtype=torch.bfloat16,
+                enabled=(args.device == "cuda"),
+            ):
+                output = model(batch)
+            loss = criterion(output, kl_target)
+            loss.backward()
+            if i % 100 == 0:
+                update_learning_rate(args, optimizer, epoch, itr=i,
+                                     itr_per_epoch=len(loader))
+            optimizer.step()
+            optimizer.zero_grad()
+            if is_gossip and not args.overlap:
+                model.transfer_params()
         if num_itr_ignore == 0:
             nn_meter.update(time.time() - nn_time)
             batch_meter.update(time.time() - batch_time)
@@ -581,6 +654,15 @@ def main(argv=None):
     best_val_prec1 = state.get("best_prec1", 0)
     is_gossip = not args.all_reduce
 
+    graphed_step = None
+    if (
+        args.hip_graph and args.device == "cuda" and is_gossip
+        and args.fused_sgd and not args.overlap
+        and getattr(model, "lazy_mixing", False) is not False
+    ):
+        graphed_step = GraphedStep(args, model, optimizer, criterion)
+        log.info("hipGraph-captured train step enabled")
+
     for epoch in range(start_epoch, args.num_epochs):
         sampler.set_epoch(epoch + args.seed * 90)
         if is_gossip:
@@ -588,7 +670,8 @@ def main(argv=None):
             model.block()
         train_epoch(args, log, model, criterion, optimizer, batch_meter,
                     data_meter, nn_meter, loader, epoch, start_itr,
-                    begin_time, args.num_itr_ignore)
+                    begin_time, args.num_itr_ignore,
+                    graphed_step=graphed_step)
         start_itr = 0
         if not args.train_fast:
             elapsed_time = time.time() - begin_time

@@ -154,7 +154,7 @@ def sgd_step_(
     params: torch.Tensor,
     grads: torch.Tensor,
     momentum_buf: torch.Tensor,
-    lr: float,
+    lr: Scalar,
     momentum: float = 0.0,
     weight_decay: float = 0.0,
     dampening: float = 0.0,
@@ -171,11 +171,12 @@ def sgd_step_(
     """
     if params.is_cuda:
         _ext_for(params).sgd_step_(
-            params, grads, momentum_buf, float(lr), float(momentum),
-            float(weight_decay), float(dampening), bool(nesterov),
-            bool(first_step),
+            params, grads, momentum_buf, _as_scalar_tensor(lr, params),
+            float(momentum), float(weight_decay), float(dampening),
+            bool(nesterov), bool(first_step),
         )
         return
+    lr = float(lr) if not isinstance(lr, torch.Tensor) else float(lr.item())
     d = grads
     if weight_decay != 0.0:
         d = d.add(params, alpha=weight_decay)

@@ -43,9 +43,9 @@ void sgp_pack_mix_bf16(float* x, unsigned short* out, const float* a,
                        int64_t n, hipStream_t stream);
 void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
                         int64_t n, hipStream_t stream);
-void sgp_sgd_step(float* p, const float* g, float* buf, double lr, double mu,
-                  double wd, double damp, bool nesterov, bool first, int64_t n,
-                  hipStream_t stream);
+void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
+                  double mu, double wd, double damp, bool nesterov,
+                  bool first, int64_t n, hipStream_t stream);
 }
 
 namespace {
@@ -100,17 +100,18 @@ void average_(torch::Tensor x, torch::Tensor y) {
               current_stream(x));
 }
 
-void sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
-               double momentum, double weight_decay, double dampening,
-               bool nesterov, bool first_step) {
+void sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
+               torch::Tensor lr, double momentum, double weight_decay,
+               double dampening, bool nesterov, bool first_step) {
   check_flat(p, "p");
   check_flat(g, "g");
   check_flat(buf, "buf");
+  check_scalar(lr, p);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == buf.numel(),
               "size mismatch");
   sgp_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
-               lr, momentum, weight_decay, dampening, nesterov, first_step,
-               p.numel(), current_stream(p));
+               lr.data_ptr<float>(), momentum, weight_decay, dampening,
+               nesterov, first_step, p.numel(), current_stream(p));
 }
 
 void pack_mix_bf16_(torch::Tensor x, torch::Tensor out, torch::Tensor a) {

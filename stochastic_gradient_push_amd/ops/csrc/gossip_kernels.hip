@@ -180,8 +180,10 @@ __global__ void k_add_scale_bf16(float* __restrict__ x,
 //   p  -= lr * d
 template <bool kMomentum, bool kNesterov, bool kFirst>
 __global__ void k_sgd(float* __restrict__ p, const float* __restrict__ g,
-                      float* __restrict__ buf, float lr, float mu, float wd,
+                      float* __restrict__ buf,
+                      const float* __restrict__ lr_ptr, float mu, float wd,
                       float damp, int64_t n4, int64_t n) {
+  const float lr = *lr_ptr;  // device-resident: live value under hipGraph replay
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (; i < n4; i += stride) {
@@ -277,16 +279,15 @@ void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
                      dim3(THREADS), 0, stream, x, r, a, n4, n);
 }
 
-void sgp_sgd_step(float* p, const float* g, float* buf, double lr, double mu,
-                  double wd, double damp, bool nesterov, bool first, int64_t n,
-                  hipStream_t stream) {
+void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
+                  double mu, double wd, double damp, bool nesterov,
+                  bool first, int64_t n, hipStream_t stream) {
   const int64_t n4 = n / 4;
   const dim3 grid(grid_for(n4 ? n4 : n));
-  const float lrf = (float)lr, muf = (float)mu, wdf = (float)wd,
-              dampf = (float)damp;
+  const float muf = (float)mu, wdf = (float)wd, dampf = (float)damp;
 #define LAUNCH(M, N, F)                                                   \
   hipLaunchKernelGGL((k_sgd<M, N, F>), grid, dim3(THREADS), 0, stream, p, \
-                     g, buf, lrf, muf, wdf, dampf, n4, n)
+                     g, buf, lr_ptr, muf, wdf, dampf, n4, n)
   if (mu != 0.0) {
     if (nesterov) {
       if (first) LAUNCH(true, true, true);

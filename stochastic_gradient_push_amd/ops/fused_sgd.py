@@ -42,6 +42,13 @@ class FusedSGD:
             torch.zeros_like(flatp.flat) if momentum != 0.0 else flatp.flat_grad
         )
         self._first_step = True
+        # device-resident lr: the step kernel reads it by pointer, so a
+        # hipGraph-captured step sees live schedule updates (sync_lr())
+        self._lr_dev = torch.tensor(
+            [float(lr)],
+            device=flatp.flat.device, dtype=torch.float32,
+        ) if flatp.flat.is_cuda else None
+        self._lr_host = float(lr)
         # torch-optim-style param_groups so LR schedules written against
         # torch.optim keep working
         self.param_groups = [
@@ -59,6 +66,15 @@ class FusedSGD:
         # flat buffers are zeroed, never detached — views stay wired
         self.flatp.zero_grad()
 
+    def sync_lr(self):
+        """Copy param_groups[0]['lr'] into the device scalar.  Called
+        automatically by step(); call it manually after LR-schedule
+        changes when the step itself is inside a captured hipGraph."""
+        lr = float(self.param_groups[0]["lr"])
+        if self._lr_dev is not None and lr != self._lr_host:
+            self._lr_dev.fill_(lr)
+            self._lr_host = lr
+
     @torch.no_grad()
     def step(self, closure=None):
         g = self.param_groups[0]
@@ -68,11 +84,12 @@ class FusedSGD:
             p0.grad.data_ptr() != self.flatp.flat_grad.data_ptr()
         ):
             self.flatp.rewire_grads()
+        self.sync_lr()
         sgd_step_(
             self.flatp.flat,
             self.flatp.flat_grad,
             self.momentum_buf,
-            lr=g["lr"],
+            lr=self._lr_dev if self._lr_dev is not None else g["lr"],
             momentum=g["momentum"],
             weight_decay=g["weight_decay"],
             dampening=g["dampening"],
