@@ -387,7 +387,10 @@ class GraphedStep:
         loss.backward()
         self.optimizer.step()
         self.optimizer.zero_grad()
-        return out, loss
+        # return detached handles: a live autograd graph held by the
+        # caller across the capture call keeps stale AccumulateGrad
+        # nodes alive and segfaults stream capture
+        return out.detach(), loss.detach()
 
     def __call__(self, batch, kl_target):
         if self.sx is None:
