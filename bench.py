@@ -66,11 +66,6 @@ def parse_args():
     return p.parse_args()
 
 
-def log(msg):
-    if int(os.environ.get("RANK", "0")) == 0:
-        print(msg, flush=True)
-
-
 def main():
     args = parse_args()
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -121,11 +116,16 @@ def main():
             )
             if world_size > 1 else None
         )
+        # OSGP: under hipGraph the overlap comes from kicking the
+        # gossip exchange BEFORE graph replay (comm rides xGMI while the
+        # captured compute runs), with the lazy sync state machine; the
+        # wrapper-level overlap hooks are only for the eager fallback.
+        graph_planned = cuda and not args.no_graph
         gdp = GossipDataParallel(
             model,
             graph=graph,
             push_sum=args.algorithm in ("sgp", "osgp"),
-            overlap=args.algorithm == "osgp",
+            overlap=(args.algorithm == "osgp" and not graph_planned),
             rank=rank if world_size > 1 else 0,
             world_size=world_size,
             gossip_dtype=(
@@ -187,12 +187,21 @@ def main():
         with torch.cuda.graph(graph, capture_error_mode="thread_local"):
             compute_step(inner)
 
-        def step():
-            graph.replay()
-            if gdp is not None:
+        if gdp is not None and args.algorithm == "osgp":
+            def step():
+                # overlap: merge previous round, kick the next exchange,
+                # then replay — gossip and compute run concurrently
                 gdp._query_gossip_queue(non_blocking=gdp.asynch)
                 gdp.transfer_params()
-            return None
+                graph.replay()
+                return None
+        else:
+            def step():
+                graph.replay()
+                if gdp is not None:
+                    gdp._query_gossip_queue(non_blocking=gdp.asynch)
+                    gdp.transfer_params()
+                return None
     else:
         def step():
             loss = compute_step(model)
@@ -222,7 +231,6 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    n_gpus = world_size if world_size > 1 else (1 if cuda else 0) or 1
     global_batch = args.batch_size * world_size
     images_per_sec = global_batch * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
