@@ -684,8 +684,6 @@ class GossipDataParallel(Module):
                         self.flatp.flat_grad, group=self.local_node_group
                     )
                 else:
-                    import functools
-
                     grads = [
                         p.grad.data
                         for p in self.module.parameters()

@@ -6,7 +6,7 @@ sends weight-``w`` shares to each active out-peer, with
 ``lo + sum_op(w_op * lo) == 1`` in the residual-adjusted form.
 """
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

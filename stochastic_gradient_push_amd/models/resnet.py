@@ -12,7 +12,7 @@ Layouts: call ``model.to(memory_format=torch.channels_last)`` on MI355X —
 MIOpen's NHWC convolutions are the fast path for bf16.
 """
 
-from typing import List, Optional, Type, Union
+from typing import List, Type, Union
 
 import torch
 import torch.nn as nn

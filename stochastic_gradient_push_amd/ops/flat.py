@@ -13,7 +13,7 @@ re-point every parameter at a view of it, so:
   SGD step for the whole model.
 """
 
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 

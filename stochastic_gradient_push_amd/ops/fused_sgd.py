@@ -12,7 +12,7 @@ Drop-in for the common SGD surface (param_groups with 'lr', zero_grad,
 step, state_dict/load_state_dict).
 """
 
-from typing import Optional
+
 
 import torch
 
