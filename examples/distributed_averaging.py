@@ -9,6 +9,11 @@ decentralized averaging (reference README.md:67-68).  Run:
 """
 
 import os
+import sys
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+)
 
 import torch
 import torch.distributed as dist
