@@ -46,6 +46,10 @@ void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
 void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
                   double mu, double wd, double damp, bool nesterov,
                   bool first, int64_t n, hipStream_t stream);
+void sgp_mfma_probe(const ushort_t* A, const ushort_t* B, float* C,
+                    hipStream_t s);
+void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                      int64_t M, int N, int K, hipStream_t s);
 }
 
 namespace {
@@ -136,6 +140,37 @@ void add_scale_bf16_(torch::Tensor x, torch::Tensor r, torch::Tensor a) {
   sgp_add_scale_bf16(x.data_ptr<float>(),
                      reinterpret_cast<const unsigned short*>(r.data_ptr()),
                      a.data_ptr<float>(), x.numel(), current_stream(x));
+}
+
+void mfma_probe(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16
+              && A.numel() == 16 * 32);
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == torch::kBFloat16
+              && B.numel() == 16 * 32);
+  TORCH_CHECK(C.is_cuda() && C.scalar_type() == torch::kFloat32
+              && C.numel() == 16 * 16);
+  sgp_mfma_probe(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                 reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                 C.data_ptr<float>(), current_stream(A));
+}
+
+void gemm_nt_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous()
+              && A.scalar_type() == torch::kBFloat16, "A must be bf16");
+  TORCH_CHECK(B.is_cuda() && B.is_contiguous()
+              && B.scalar_type() == torch::kBFloat16, "B must be bf16");
+  TORCH_CHECK(C.is_cuda() && C.is_contiguous()
+              && C.scalar_type() == torch::kBFloat16, "C must be bf16");
+  const int64_t M = A.size(0);
+  const int64_t K = A.size(1);
+  const int64_t N = B.size(0);
+  TORCH_CHECK(B.size(1) == K && C.size(0) == M && C.size(1) == N,
+              "shape mismatch");
+  TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+  sgp_gemm_nt_bf16(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                   reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                   reinterpret_cast<ushort_t*>(C.data_ptr()), M, (int)N,
+                   (int)K, current_stream(A));
 }
 
 // ---------------------------------------------------------------- BN ops
@@ -283,6 +318,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_mix_", &pack_mix_, "x *= a; out = x");
   m.def("average_", &average_, "x = (x + y) / 2");
   m.def("sgd_step_", &sgd_step_, "fused momentum-SGD step");
+  m.def("mfma_probe", &mfma_probe, "single 16x16x32 MFMA layout probe");
+  m.def("gemm_nt_bf16", &gemm_nt_bf16,
+        "C[M,N] = A[M,K] @ B[N,K]^T, bf16 MFMA, fp32 accumulate");
   m.def("pack_mix_bf16_", &pack_mix_bf16_,
         "x *= a; out_bf16 = bf16(x) (wire-format pack)");
   m.def("add_scale_bf16_", &add_scale_bf16_,
