@@ -50,6 +50,8 @@ void sgp_mfma_probe(const ushort_t* A, const ushort_t* B, float* C,
                     hipStream_t s);
 void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
                       int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s);
 }
 
 namespace {
@@ -152,6 +154,27 @@ void mfma_probe(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
   sgp_mfma_probe(reinterpret_cast<const ushort_t*>(A.data_ptr()),
                  reinterpret_cast<const ushort_t*>(B.data_ptr()),
                  C.data_ptr<float>(), current_stream(A));
+}
+
+void gemm_nt_check(const torch::Tensor& A, const torch::Tensor& B,
+                   const torch::Tensor& C) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous()
+              && A.scalar_type() == torch::kBFloat16, "A must be bf16");
+  TORCH_CHECK(B.is_cuda() && B.is_contiguous()
+              && B.scalar_type() == torch::kBFloat16, "B must be bf16");
+  TORCH_CHECK(C.is_cuda() && C.is_contiguous()
+              && C.scalar_type() == torch::kBFloat16, "C must be bf16");
+  TORCH_CHECK(B.size(1) == A.size(1) && C.size(0) == A.size(0)
+              && C.size(1) == B.size(0), "shape mismatch");
+  TORCH_CHECK(A.size(1) % 32 == 0, "K must be a multiple of 32");
+}
+
+void gemm_nt_bf16_v2(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  gemm_nt_check(A, B, C);
+  sgp_gemm_nt_bf16_v2(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), current_stream(A));
 }
 
 void gemm_nt_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
@@ -321,6 +344,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 MFMA layout probe");
   m.def("gemm_nt_bf16", &gemm_nt_bf16,
         "C[M,N] = A[M,K] @ B[N,K]^T, bf16 MFMA, fp32 accumulate");
+  m.def("gemm_nt_bf16_v2", &gemm_nt_bf16_v2,
+        "pipelined (register-staged double-buffer) variant");
   m.def("pack_mix_bf16_", &pack_mix_bf16_,
         "x *= a; out_bf16 = bf16(x) (wire-format pack)");
   m.def("add_scale_bf16_", &add_scale_bf16_,

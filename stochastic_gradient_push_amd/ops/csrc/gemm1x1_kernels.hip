@@ -179,6 +179,114 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16(
   }
 }
 
+// ------------------------------------------------- gemm v2 (pipelined)
+// Same geometry, register-staged double buffering (guide T14 shape):
+// global loads for tile t+1 are issued right after the barrier, compute
+// runs on tile t from LDS, and the staged registers are written to the
+// other LDS buffer after the next barrier — global latency hides behind
+// the MFMA block.
+__global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    ushort_t* __restrict__ C, int64_t M, int N, int K) {
+  __shared__ ushort_t As[2][BM * LDS_STRIDE];
+  __shared__ ushort_t Bs[2][BN * LDS_STRIDE];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;
+
+  // each thread stages 2 segments x (A,B): seg -> (row, col8)
+  const int seg0 = tid, seg1 = tid + 256;
+  const int r0 = seg0 >> 2, c0 = (seg0 & 3) * 8;
+  const int r1 = seg1 >> 2, c1 = (seg1 & 3) * 8;
+
+  const int n_tiles = (N + BN - 1) / BN;
+  const int64_t m_tiles = (M + BM - 1) / BM;
+  const int64_t total_tiles = m_tiles * n_tiles;
+  const int KT = K / BK;
+
+  for (int64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+    const int64_t tm = (tile / n_tiles) * BM;
+    const int tn = (int)(tile % n_tiles) * BN;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    ushort_t ra0[8], ra1[8], rb0[8], rb1[8];
+
+#define LOAD_TILE(k0)                                                      \
+  do {                                                                     \
+    const int64_t ga0 = tm + r0, ga1 = tm + r1;                            \
+    const int gb0 = tn + r0, gb1 = tn + r1;                                \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                        \
+      ra0[j] = (ga0 < M) ? A[ga0 * K + (k0) + c0 + j] : (ushort_t)0;       \
+      ra1[j] = (ga1 < M) ? A[ga1 * K + (k0) + c1 + j] : (ushort_t)0;       \
+      rb0[j] = (gb0 < N) ? B[(int64_t)gb0 * K + (k0) + c0 + j] : (ushort_t)0; \
+      rb1[j] = (gb1 < N) ? B[(int64_t)gb1 * K + (k0) + c1 + j] : (ushort_t)0; \
+    }                                                                      \
+  } while (0)
+
+#define WRITE_TILE(buf)                                                    \
+  do {                                                                     \
+    *reinterpret_cast<bf16x8*>(As[buf] + r0 * LDS_STRIDE + c0) =           \
+        *reinterpret_cast<bf16x8*>(ra0);                                   \
+    *reinterpret_cast<bf16x8*>(As[buf] + r1 * LDS_STRIDE + c1) =           \
+        *reinterpret_cast<bf16x8*>(ra1);                                   \
+    *reinterpret_cast<bf16x8*>(Bs[buf] + r0 * LDS_STRIDE + c0) =           \
+        *reinterpret_cast<bf16x8*>(rb0);                                   \
+    *reinterpret_cast<bf16x8*>(Bs[buf] + r1 * LDS_STRIDE + c1) =           \
+        *reinterpret_cast<bf16x8*>(rb1);                                   \
+  } while (0)
+
+    LOAD_TILE(0);
+    WRITE_TILE(0);
+
+    for (int kt = 0; kt < KT; ++kt) {
+      __syncthreads();
+      const int buf = kt & 1;
+      if (kt + 1 < KT) LOAD_TILE((kt + 1) * BK);  // issue early
+
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            As[buf] + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
+        bfrag[i] = *reinterpret_cast<const bf16x8*>(
+            Bs[buf] + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+
+      __syncthreads();
+      if (kt + 1 < KT) WRITE_TILE(buf ^ 1);
+    }
+#undef LOAD_TILE
+#undef WRITE_TILE
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gc = tn + wn + j * 16 + (lane & 15);
+          if (gr < M && gc < N) C[gr * N + gc] = f2b(acc[i][j][r]);
+        }
+  }
+}
+
 inline int gemm_grid(int64_t M, int N) {
   int64_t tiles = ((M + BM - 1) / BM) * (int64_t)((N + BN - 1) / BN);
   if (tiles > 16384) tiles = 16384;
@@ -199,6 +307,12 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
                       int64_t M, int N, int K, hipStream_t s) {
   hipLaunchKernelGGL(k_gemm_nt_bf16, dim3(gemm_grid(M, N)), dim3(256), 0, s,
                      A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s) {
+  hipLaunchKernelGGL(k_gemm_nt_bf16_v2, dim3(gemm_grid(M, N)), dim3(256), 0,
+                     s, A, B, C, M, N, K);
 }
 
 }  // extern "C"

@@ -99,3 +99,41 @@ def test_gemm_nt_throughput_readout():
         f"{flops / t_lib / 1e12:.1f} TF ({t_lib * 1e6:.0f} us)"
     )
     assert t_ours < 1.0  # sanity only
+
+
+@pytest.mark.parametrize("shape", [(25088, 512, 512), (1568, 2048, 512),
+                                   (130, 72, 64)])
+def test_gemm_nt_v2_matches_v1(shape):
+    M, N, K = shape
+    torch.manual_seed(2)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C2 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v2(A, B, C2)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C2)
+
+
+def test_gemm_v2_throughput_readout():
+    M, N, K = 25088, 512, 512
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t1 = t(lambda: ext().gemm_nt_bf16(A, B, C))
+    t2 = t(lambda: ext().gemm_nt_bf16_v2(A, B, C))
+    print(f"\n[gemm v1 vs v2] {flops / t1 / 1e12:.1f} TF vs "
+          f"{flops / t2 / 1e12:.1f} TF")
