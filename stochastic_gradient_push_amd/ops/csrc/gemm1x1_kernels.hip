@@ -108,41 +108,34 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16(
       // stage A/B tiles: 128 rows x 32 cols, 16 B (8 bf16) per access,
       // 2 segments per thread per operand
       __syncthreads();
+      const bool full = (tm + BM <= M) && (tn + BN <= N);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         const int seg = tid + s * 256;     // 0..511
         const int row = seg >> 2;          // /4
         const int c8 = (seg & 3) * 8;
-        // A
-        {
-          const int64_t gr = tm + row;
-          ushort_t tmp[8];
-          if (gr < M) {
-            const ushort_t* src = A + gr * K + k0 + c8;
+        const int64_t gra = tm + row;
+        const int grb = tn + row;
+        bf16x8 va, vb;
+        if (full) {
+          // one 16-B vector load per operand (scalar bf16 loads are
+          // ~2.5x slower; per-element guards serialize — guide traps)
+          va = *reinterpret_cast<const bf16x8*>(A + gra * K + k0 + c8);
+          vb = *reinterpret_cast<const bf16x8*>(
+              B + (int64_t)grb * K + k0 + c8);
+        } else {
+          ushort_t ta[8], tb[8];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = src[j];
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = 0;
+          for (int j = 0; j < 8; ++j) {
+            ta[j] = (gra < M) ? A[gra * K + k0 + c8 + j] : (ushort_t)0;
+            tb[j] = (grb < N)
+                ? B[(int64_t)grb * K + k0 + c8 + j] : (ushort_t)0;
           }
-          *reinterpret_cast<bf16x8*>(As + row * LDS_STRIDE + c8) =
-              *reinterpret_cast<bf16x8*>(tmp);
+          va = *reinterpret_cast<bf16x8*>(ta);
+          vb = *reinterpret_cast<bf16x8*>(tb);
         }
-        // B
-        {
-          const int gr = tn + row;
-          ushort_t tmp[8];
-          if (gr < N) {
-            const ushort_t* src = B + (int64_t)gr * K + k0 + c8;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = src[j];
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = 0;
-          }
-          *reinterpret_cast<bf16x8*>(Bs + row * LDS_STRIDE + c8) =
-              *reinterpret_cast<bf16x8*>(tmp);
-        }
+        *reinterpret_cast<bf16x8*>(As + row * LDS_STRIDE + c8) = va;
+        *reinterpret_cast<bf16x8*>(Bs + row * LDS_STRIDE + c8) = vb;
       }
       __syncthreads();
 
@@ -219,30 +212,43 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
 #pragma unroll
       for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    ushort_t ra0[8], ra1[8], rb0[8], rb1[8];
+    bf16x8 ra0, ra1, rb0, rb1;
+    const bool full = (tm + BM <= M) && (tn + BN <= N);
 
 #define LOAD_TILE(k0)                                                      \
   do {                                                                     \
     const int64_t ga0 = tm + r0, ga1 = tm + r1;                            \
     const int gb0 = tn + r0, gb1 = tn + r1;                                \
-    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                        \
-      ra0[j] = (ga0 < M) ? A[ga0 * K + (k0) + c0 + j] : (ushort_t)0;       \
-      ra1[j] = (ga1 < M) ? A[ga1 * K + (k0) + c1 + j] : (ushort_t)0;       \
-      rb0[j] = (gb0 < N) ? B[(int64_t)gb0 * K + (k0) + c0 + j] : (ushort_t)0; \
-      rb1[j] = (gb1 < N) ? B[(int64_t)gb1 * K + (k0) + c1 + j] : (ushort_t)0; \
+    if (full) {                                                            \
+      ra0 = *reinterpret_cast<const bf16x8*>(A + ga0 * K + (k0) + c0);     \
+      ra1 = *reinterpret_cast<const bf16x8*>(A + ga1 * K + (k0) + c1);     \
+      rb0 = *reinterpret_cast<const bf16x8*>(                              \
+          B + (int64_t)gb0 * K + (k0) + c0);                               \
+      rb1 = *reinterpret_cast<const bf16x8*>(                              \
+          B + (int64_t)gb1 * K + (k0) + c1);                               \
+    } else {                                                               \
+      ushort_t t0[8], t1[8], t2[8], t3[8];                                 \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
+        t0[j] = (ga0 < M) ? A[ga0 * K + (k0) + c0 + j] : (ushort_t)0;      \
+        t1[j] = (ga1 < M) ? A[ga1 * K + (k0) + c1 + j] : (ushort_t)0;      \
+        t2[j] = (gb0 < N) ? B[(int64_t)gb0 * K + (k0) + c0 + j]            \
+                          : (ushort_t)0;                                   \
+        t3[j] = (gb1 < N) ? B[(int64_t)gb1 * K + (k0) + c1 + j]            \
+                          : (ushort_t)0;                                   \
+      }                                                                    \
+      ra0 = *reinterpret_cast<bf16x8*>(t0);                                \
+      ra1 = *reinterpret_cast<bf16x8*>(t1);                                \
+      rb0 = *reinterpret_cast<bf16x8*>(t2);                                \
+      rb1 = *reinterpret_cast<bf16x8*>(t3);                                \
     }                                                                      \
   } while (0)
 
 #define WRITE_TILE(buf)                                                    \
   do {                                                                     \
-    *reinterpret_cast<bf16x8*>(As[buf] + r0 * LDS_STRIDE + c0) =           \
-        *reinterpret_cast<bf16x8*>(ra0);                                   \
-    *reinterpret_cast<bf16x8*>(As[buf] + r1 * LDS_STRIDE + c1) =           \
-        *reinterpret_cast<bf16x8*>(ra1);                                   \
-    *reinterpret_cast<bf16x8*>(Bs[buf] + r0 * LDS_STRIDE + c0) =           \
-        *reinterpret_cast<bf16x8*>(rb0);                                   \
-    *reinterpret_cast<bf16x8*>(Bs[buf] + r1 * LDS_STRIDE + c1) =           \
-        *reinterpret_cast<bf16x8*>(rb1);                                   \
+    *reinterpret_cast<bf16x8*>(As[buf] + r0 * LDS_STRIDE + c0) = ra0;      \
+    *reinterpret_cast<bf16x8*>(As[buf] + r1 * LDS_STRIDE + c1) = ra1;      \
+    *reinterpret_cast<bf16x8*>(Bs[buf] + r0 * LDS_STRIDE + c0) = rb0;      \
+    *reinterpret_cast<bf16x8*>(Bs[buf] + r1 * LDS_STRIDE + c1) = rb1;      \
   } while (0)
 
     LOAD_TILE(0);
