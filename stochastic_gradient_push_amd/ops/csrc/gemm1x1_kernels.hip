@@ -178,11 +178,14 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16(
 // runs on tile t from LDS, and the staged registers are written to the
 // other LDS buffer after the next barrier — global latency hides behind
 // the MFMA block.
+template <int BKT>
 __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     ushort_t* __restrict__ C, int64_t M, int N, int K) {
-  __shared__ ushort_t As[2][BM * LDS_STRIDE];
-  __shared__ ushort_t Bs[2][BN * LDS_STRIDE];
+  constexpr int STRIDE = BKT + 8;          // +16B pad per row
+  constexpr int SEGS = BM * BKT / 8 / 256; // 16B segments per thread
+  __shared__ ushort_t As[2][BM * STRIDE];
+  __shared__ ushort_t Bs[2][BN * STRIDE];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -192,15 +195,20 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
   const int frow = lane & 15;
   const int fk0 = (lane >> 4) * 8;
 
-  // each thread stages 2 segments x (A,B): seg -> (row, col8)
-  const int seg0 = tid, seg1 = tid + 256;
-  const int r0 = seg0 >> 2, c0 = (seg0 & 3) * 8;
-  const int r1 = seg1 >> 2, c1 = (seg1 & 3) * 8;
+  // each thread stages SEGS segments x (A,B): seg -> (row, col8)
+  constexpr int SEG_PER_ROW = BKT / 8;
+  int rr[SEGS], cc[SEGS];
+#pragma unroll
+  for (int s = 0; s < SEGS; ++s) {
+    const int seg = tid + s * 256;
+    rr[s] = seg / SEG_PER_ROW;
+    cc[s] = (seg % SEG_PER_ROW) * 8;
+  }
 
   const int n_tiles = (N + BN - 1) / BN;
   const int64_t m_tiles = (M + BM - 1) / BM;
   const int64_t total_tiles = m_tiles * n_tiles;
-  const int KT = K / BK;
+  const int KT = K / BKT;
 
   for (int64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
     const int64_t tm = (tile / n_tiles) * BM;
@@ -212,43 +220,42 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
 #pragma unroll
       for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    bf16x8 ra0, ra1, rb0, rb1;
+    bf16x8 ra[SEGS], rb[SEGS];
     const bool full = (tm + BM <= M) && (tn + BN <= N);
 
 #define LOAD_TILE(k0)                                                      \
   do {                                                                     \
-    const int64_t ga0 = tm + r0, ga1 = tm + r1;                            \
-    const int gb0 = tn + r0, gb1 = tn + r1;                                \
     if (full) {                                                            \
-      ra0 = *reinterpret_cast<const bf16x8*>(A + ga0 * K + (k0) + c0);     \
-      ra1 = *reinterpret_cast<const bf16x8*>(A + ga1 * K + (k0) + c1);     \
-      rb0 = *reinterpret_cast<const bf16x8*>(                              \
-          B + (int64_t)gb0 * K + (k0) + c0);                               \
-      rb1 = *reinterpret_cast<const bf16x8*>(                              \
-          B + (int64_t)gb1 * K + (k0) + c1);                               \
-    } else {                                                               \
-      ushort_t t0[8], t1[8], t2[8], t3[8];                                 \
-      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
-        t0[j] = (ga0 < M) ? A[ga0 * K + (k0) + c0 + j] : (ushort_t)0;      \
-        t1[j] = (ga1 < M) ? A[ga1 * K + (k0) + c1 + j] : (ushort_t)0;      \
-        t2[j] = (gb0 < N) ? B[(int64_t)gb0 * K + (k0) + c0 + j]            \
-                          : (ushort_t)0;                                   \
-        t3[j] = (gb1 < N) ? B[(int64_t)gb1 * K + (k0) + c1 + j]            \
-                          : (ushort_t)0;                                   \
+      _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                   \
+        ra[s] = *reinterpret_cast<const bf16x8*>(                          \
+            A + (tm + rr[s]) * K + (k0) + cc[s]);                          \
+        rb[s] = *reinterpret_cast<const bf16x8*>(                          \
+            B + (int64_t)(tn + rr[s]) * K + (k0) + cc[s]);                 \
       }                                                                    \
-      ra0 = *reinterpret_cast<bf16x8*>(t0);                                \
-      ra1 = *reinterpret_cast<bf16x8*>(t1);                                \
-      rb0 = *reinterpret_cast<bf16x8*>(t2);                                \
-      rb1 = *reinterpret_cast<bf16x8*>(t3);                                \
+    } else {                                                               \
+      _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                   \
+        const int64_t ga = tm + rr[s];                                     \
+        const int gb = tn + rr[s];                                         \
+        ushort_t ta[8], tb[8];                                             \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j) {                    \
+          ta[j] = (ga < M) ? A[ga * K + (k0) + cc[s] + j] : (ushort_t)0;   \
+          tb[j] = (gb < N) ? B[(int64_t)gb * K + (k0) + cc[s] + j]         \
+                           : (ushort_t)0;                                  \
+        }                                                                  \
+        ra[s] = *reinterpret_cast<bf16x8*>(ta);                            \
+        rb[s] = *reinterpret_cast<bf16x8*>(tb);                            \
+      }                                                                    \
     }                                                                      \
   } while (0)
 
 #define WRITE_TILE(buf)                                                    \
   do {                                                                     \
-    *reinterpret_cast<bf16x8*>(As[buf] + r0 * LDS_STRIDE + c0) = ra0;      \
-    *reinterpret_cast<bf16x8*>(As[buf] + r1 * LDS_STRIDE + c1) = ra1;      \
-    *reinterpret_cast<bf16x8*>(Bs[buf] + r0 * LDS_STRIDE + c0) = rb0;      \
-    *reinterpret_cast<bf16x8*>(Bs[buf] + r1 * LDS_STRIDE + c1) = rb1;      \
+    _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                     \
+      *reinterpret_cast<bf16x8*>(As[buf] + rr[s] * STRIDE + cc[s]) =       \
+          ra[s];                                                           \
+      *reinterpret_cast<bf16x8*>(Bs[buf] + rr[s] * STRIDE + cc[s]) =       \
+          rb[s];                                                           \
+    }                                                                      \
   } while (0)
 
     LOAD_TILE(0);
@@ -257,22 +264,25 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
     for (int kt = 0; kt < KT; ++kt) {
       __syncthreads();
       const int buf = kt & 1;
-      if (kt + 1 < KT) LOAD_TILE((kt + 1) * BK);  // issue early
+      if (kt + 1 < KT) LOAD_TILE((int64_t)(kt + 1) * BKT);  // issue early
 
-      bf16x8 afrag[4], bfrag[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        afrag[i] = *reinterpret_cast<const bf16x8*>(
-            As[buf] + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
-        bfrag[i] = *reinterpret_cast<const bf16x8*>(
-            Bs[buf] + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
+      for (int ks = 0; ks < BKT / 32; ++ks) {
+        bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          afrag[i] = *reinterpret_cast<const bf16x8*>(
+              As[buf] + (wm + i * 16 + frow) * STRIDE + ks * 32 + fk0);
+          bfrag[i] = *reinterpret_cast<const bf16x8*>(
+              Bs[buf] + (wn + i * 16 + frow) * STRIDE + ks * 32 + fk0);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
       }
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
 
       __syncthreads();
       if (kt + 1 < KT) WRITE_TILE(buf ^ 1);
@@ -317,8 +327,12 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
 
 void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s) {
-  hipLaunchKernelGGL(k_gemm_nt_bf16_v2, dim3(gemm_grid(M, N)), dim3(256), 0,
-                     s, A, B, C, M, N, K);
+  if (K % 64 == 0)
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v2<64>), dim3(gemm_grid(M, N)),
+                       dim3(256), 0, s, A, B, C, M, N, K);
+  else
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v2<32>), dim3(gemm_grid(M, N)),
+                       dim3(256), 0, s, A, B, C, M, N, K);
 }
 
 }  // extern "C"

@@ -137,3 +137,47 @@ def _heartbeat(rank, world_size):
 
 def test_gossip_heartbeat_timeout():
     run_dist(_heartbeat, world_size=2)
+
+
+def _runtime_ppi_change(rank, world_size):
+    """update_gossiper('peers_per_itr', v) under the gossip lock changes
+    the active out-degree mid-training (reference distributed.py:197-207,
+    driven by the trainer's peers-per-itr schedule)."""
+    from stochastic_gradient_push_amd import (
+        GossipDataParallel,
+        NPeerDynamicDirectedExponentialGraph,
+    )
+
+    model = tiny_model(seed=rank)
+    graph = NPeerDynamicDirectedExponentialGraph(
+        rank, world_size, peers_per_itr=2
+    )
+    gdp = GossipDataParallel(model, graph=graph, push_sum=True)
+    opt = torch.optim.SGD(gdp.parameters(), lr=0.0)
+    x = torch.randn(2, 6)
+    y = torch.randn(2, 4)
+    gdp.train()
+
+    def step():
+        loss = ((gdp(x) - y) ** 2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        gdp.transfer_params()
+
+    for _ in range(3):
+        step()
+    gdp.sync_comms()
+    gdp.update_gossiper("peers_per_itr", 1)
+    gossiper = list(gdp.dist_config["gossipers"].values())[0]
+    assert gossiper.peers_per_itr == 1
+    for _ in range(3):
+        step()
+    gdp.sync_comms()
+    gdp.unbias()
+    assert torch.isfinite(gdp.flatp.flat).all()
+    gdp.shutdown()
+
+
+def test_runtime_peers_per_itr_change():
+    run_dist(_runtime_ppi_change, world_size=4)
