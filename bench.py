@@ -178,15 +178,22 @@ def main():
     )
     if use_graph:
         inner = gdp.module if gdp is not None else model
-        for _ in range(3):
-            compute_step(inner)
-        torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
-        # thread_local: the (idle) gossip thread and NCCL watchdog must
-        # not poison a global-mode stream capture at world_size > 1
-        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
-            compute_step(inner)
+        try:
+            for _ in range(3):
+                compute_step(inner)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            # thread_local: the (idle) gossip thread and NCCL watchdog
+            # must not poison a global-mode capture at world_size > 1
+            with torch.cuda.graph(graph, capture_error_mode="thread_local"):
+                compute_step(inner)
+        except Exception as e:  # insurance: never lose the bench to capture
+            print(f"[bench] hipGraph capture failed ({e}); eager fallback",
+                  flush=True)
+            torch.cuda.synchronize()
+            use_graph = False
 
+    if use_graph:
         if gdp is not None and args.algorithm == "osgp":
             def step():
                 # overlap: merge previous round, kick the next exchange,
