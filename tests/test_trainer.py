@@ -169,3 +169,30 @@ def test_adpsgd_cli_two_rank(tmp_path):
         script="gossip_sgd_adpsgd.py", nprocs=2,
     )
     assert os.path.exists(f"{tmp_path}/ckpt/out_r0_n2.csv")
+
+
+def test_bench_json_contract(tmp_path):
+    """The driver depends on bench.py's JSON line: validate schema."""
+    import json
+
+    env = dict(os.environ)
+    r = subprocess.run(
+        [
+            sys.executable, os.path.join(REPO, "bench.py"),
+            "--device", "cpu", "--steps", "2", "--warmup", "1",
+            "--batch-size", "2", "--model", "resnet18", "--dtype", "fp32",
+        ],
+        env=env, check=True, timeout=300, cwd=str(tmp_path),
+        capture_output=True, text=True,
+    )
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in d, key
+    assert d["metric"] == "images/sec"
+    assert d["scaling"] == "weak"
+    assert d["data"] == "synthetic"
+    assert d["config"]["model"] == "resnet18"
+    assert d["value"] > 0
