@@ -52,6 +52,8 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
                       int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s);
 }
 
 namespace {
@@ -172,6 +174,15 @@ void gemm_nt_check(const torch::Tensor& A, const torch::Tensor& B,
 void gemm_nt_bf16_v2(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
   gemm_nt_check(A, B, C);
   sgp_gemm_nt_bf16_v2(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), current_stream(A));
+}
+
+void gemm_nt_bf16_v3(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(1) % 64 == 0, "v3 requires K %% 64 == 0");
+  sgp_gemm_nt_bf16_v3(reinterpret_cast<const ushort_t*>(A.data_ptr()),
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), current_stream(A));
@@ -346,6 +357,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "C[M,N] = A[M,K] @ B[N,K]^T, bf16 MFMA, fp32 accumulate");
   m.def("gemm_nt_bf16_v2", &gemm_nt_bf16_v2,
         "pipelined (register-staged double-buffer) variant");
+  m.def("gemm_nt_bf16_v3", &gemm_nt_bf16_v3,
+        "global_load_lds + st_16x32 swizzle variant (K % 64 == 0)");
   m.def("pack_mix_bf16_", &pack_mix_bf16_,
         "x *= a; out_bf16 = bf16(x) (wire-format pack)");
   m.def("add_scale_bf16_", &add_scale_bf16_,

@@ -303,6 +303,188 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v2(
   }
 }
 
+// --------------------------------------------- gemm v3 (glds + swizzle)
+// BK=64 only.  Full tiles are staged with __builtin_amdgcn_global_load_lds
+// (16-B direct-to-LDS DMA, no staging VGPRs / ds_write pass) into an
+// UNPADDED [128][64] bf16 image with the st_16x32 XOR swizzle
+// (byte ^= ((byte>>9)&1)<<5): glds writes are lane-linear, so the
+// swizzle is applied to the per-lane GLOBAL source address, and the
+// fragment ds_read_b128 uses the matching swizzled offset (guide
+// recipe; linear layout would be 8-way bank-conflicted).  Tail tiles
+// (M or N remainder) fall back to guarded register staging.
+#define BK3 64
+#define TILE_BYTES (BM * BK3 * 2)  // 16 KiB per operand
+
+__device__ __forceinline__ int swz(int row, int cb) {
+  // byte-offset swizzle within the [128][128B] tile image
+  return cb ^ (((row >> 2) & 1) << 5);
+}
+
+__global__ __launch_bounds__(256) void k_gemm_nt_bf16_v3(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    ushort_t* __restrict__ C, int64_t M, int N, int K) {
+  // one shared object only (a second one forces vmcnt(0) before every
+  // ds_read of a glds pipeline — guide trap 4a)
+  __shared__ ushort_t lds[2 * 2 * BM * BK3];  // [buf][A,B][128][64]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;  // fragment k-offset in BYTES
+
+  // glds geometry: each wave stages 4 chunks of 1 KiB per operand;
+  // chunk c covers rows [c*8, c*8+8), lane l -> row c*8 + l/8,
+  // byte col (l%8)*16 within the 128-B row
+  const int g_row_in_chunk = lane >> 3;
+  const int g_cb = (lane & 7) * 16;
+
+  const int n_tiles = (N + BN - 1) / BN;
+  const int64_t m_tiles = (M + BM - 1) / BM;
+  const int64_t total_tiles = m_tiles * n_tiles;
+  const int KT = K / BK3;
+
+  for (int64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+    const int64_t tm = (tile / n_tiles) * BM;
+    const int tn = (int)(tile % n_tiles) * BN;
+    const bool full = (tm + BM <= M) && (tn + BN <= N);
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    if (full) {
+      // ---- glds pipeline (2 LDS buffers, vmcnt drained by barrier) ----
+#define GLDS_TILE(buf, k0)                                                 \
+  do {                                                                     \
+    _Pragma("unroll") for (int c = 0; c < 4; ++c) {                        \
+      const int row = (wave * 4 + c) * 8 + g_row_in_chunk;                 \
+      const int scb = swz(row, g_cb);                                      \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              A + (tm + row) * K + (k0) + scb / 2),                        \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + (buf) * 2 * BM * BK3 + (wave * 4 + c) * 512),          \
+          16, 0, 0);                                                       \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              B + (int64_t)(tn + row) * K + (k0) + scb / 2),               \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + ((buf) * 2 + 1) * BM * BK3 + (wave * 4 + c) * 512),    \
+          16, 0, 0);                                                       \
+    }                                                                      \
+  } while (0)
+
+      GLDS_TILE(0, 0);
+      for (int kt = 0; kt < KT; ++kt) {
+        const int buf = kt & 1;
+        __syncthreads();  // drains the in-flight glds for buf (vmcnt(0))
+        if (kt + 1 < KT) GLDS_TILE(buf ^ 1, (int64_t)(kt + 1) * BK3);
+
+        const ushort_t* As_ = lds + buf * 2 * BM * BK3;
+        const ushort_t* Bs_ = lds + (buf * 2 + 1) * BM * BK3;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int ra = wm + i * 16 + frow;
+            const int rb = wn + i * 16 + frow;
+            const int cb = ks * 64 + fkb;
+            afrag[i] = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(As_) + ra * 128
+                + swz(ra, cb));
+            bfrag[i] = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(Bs_) + rb * 128
+                + swz(rb, cb));
+          }
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+        }
+      }
+#undef GLDS_TILE
+    } else {
+      // ---- tail tiles: guarded register staging into buffer 0, linear
+      // layout with the same swizzled addressing (write side applies
+      // the swizzle so the read side is uniform) ----
+      for (int kt = 0; kt < KT; ++kt) {
+        __syncthreads();
+#pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          const int seg = tid + s * 256;   // 512 segs x 16 B = A+B tile
+          const int half = seg >> 8;       // 0: A, 1: B
+          const int idx = seg & 255;       // 128 rows x 2 halves... rows*2
+          const int row = idx >> 1;
+          const int cb = (idx & 1) * 64 + 0;  // two 64-B pieces per row
+          // stage 64 B per segment as 4 x 16 B guarded pieces
+#pragma unroll
+          for (int p = 0; p < 4; ++p) {
+            const int cbb = cb + p * 16;
+            ushort_t tmp[8];
+            const int64_t gr = (half ? (int64_t)tn : tm) + row;
+            const int64_t lim = half ? (int64_t)N : M;
+            const ushort_t* base = half ? B : A;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              tmp[j] = (gr < lim)
+                  ? base[gr * K + (int64_t)kt * BK3 + cbb / 2 + j]
+                  : (ushort_t)0;
+            }
+            *reinterpret_cast<bf16x8*>(
+                reinterpret_cast<char*>(lds) + half * BM * BK3 * 2
+                + row * 128 + swz(row, cbb)) =
+                *reinterpret_cast<bf16x8*>(tmp);
+          }
+        }
+        __syncthreads();
+        const ushort_t* As_ = lds;
+        const ushort_t* Bs_ = lds + BM * BK3;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int ra = wm + i * 16 + frow;
+            const int rb = wn + i * 16 + frow;
+            const int cb = ks * 64 + fkb;
+            afrag[i] = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(As_) + ra * 128
+                + swz(ra, cb));
+            bfrag[i] = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(Bs_) + rb * 128
+                + swz(rb, cb));
+          }
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+        }
+      }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gc = tn + wn + j * 16 + (lane & 15);
+          if (gr < M && gc < N) C[gr * N + gc] = f2b(acc[i][j][r]);
+        }
+  }
+}
+
 inline int gemm_grid(int64_t M, int N) {
   int64_t tiles = ((M + BM - 1) / BM) * (int64_t)((N + BN - 1) / BN);
   if (tiles > 16384) tiles = 16384;
@@ -323,6 +505,12 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
                       int64_t M, int N, int K, hipStream_t s) {
   hipLaunchKernelGGL(k_gemm_nt_bf16, dim3(gemm_grid(M, N)), dim3(256), 0, s,
                      A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s) {
+  hipLaunchKernelGGL(k_gemm_nt_bf16_v3, dim3(gemm_grid(M, N)), dim3(256), 0,
+                     s, A, B, C, M, N, K);
 }
 
 void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,

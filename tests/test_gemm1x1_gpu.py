@@ -137,3 +137,43 @@ def test_gemm_v2_throughput_readout():
     t2 = t(lambda: ext().gemm_nt_bf16_v2(A, B, C))
     print(f"\n[gemm v1 vs v2] {flops / t1 / 1e12:.1f} TF vs "
           f"{flops / t2 / 1e12:.1f} TF")
+
+
+@pytest.mark.parametrize("shape", [(25088, 512, 512), (1568, 2048, 512),
+                                   (256, 64, 64), (130, 72, 64)])
+def test_gemm_nt_v3_matches_v1(shape):
+    M, N, K = shape
+    torch.manual_seed(3)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C3 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v3(A, B, C3)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C3), (
+        (C1.float() - C3.float()).abs().max().item()
+    )
+
+
+def test_gemm_v3_throughput_readout():
+    M, N, K = 25088, 512, 512
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t2 = t(lambda: ext().gemm_nt_bf16_v2(A, B, C))
+    t3 = t(lambda: ext().gemm_nt_bf16_v3(A, B, C))
+    print(f"\n[gemm v2 vs v3-glds] {flops / t2 / 1e12:.1f} TF vs "
+          f"{flops / t3 / 1e12:.1f} TF")
