@@ -54,6 +54,8 @@ void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s);
 }
 
 namespace {
@@ -183,6 +185,17 @@ void gemm_nt_bf16_v3(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
   gemm_nt_check(A, B, C);
   TORCH_CHECK(A.size(1) % 64 == 0, "v3 requires K %% 64 == 0");
   sgp_gemm_nt_bf16_v3(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), current_stream(A));
+}
+
+void gemm_nt_bf16_v4(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(0) % 128 == 0 && B.size(0) % 128 == 0
+              && A.size(1) % 64 == 0,
+              "v4 requires M,N %% 128 == 0 and K %% 64 == 0");
+  sgp_gemm_nt_bf16_v4(reinterpret_cast<const ushort_t*>(A.data_ptr()),
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), current_stream(A));
@@ -359,6 +372,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "pipelined (register-staged double-buffer) variant");
   m.def("gemm_nt_bf16_v3", &gemm_nt_bf16_v3,
         "global_load_lds + st_16x32 swizzle variant (K % 64 == 0)");
+  m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
+        "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
   m.def("pack_mix_bf16_", &pack_mix_bf16_,
         "x *= a; out_bf16 = bf16(x) (wire-format pack)");
   m.def("add_scale_bf16_", &add_scale_bf16_,

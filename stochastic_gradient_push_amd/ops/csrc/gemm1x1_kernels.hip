@@ -485,6 +485,120 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v3(
   }
 }
 
+// ----------------------------------- gemm v4 (3-buffer glds, raw barrier)
+// Full-tile-only variant: keeps one tile of glds in flight ACROSS the
+// barrier (guide: __syncthreads() drains glds via its vmcnt(0) fence —
+// the ~20% stall of the 2-buffer structure; raw s_barrier + counted
+// s_waitcnt vmcnt(N) leaves the prefetch in flight).  Each wave issues
+// 8 glds per tile; vmcnt(8) at the top of the loop means "my tile-k
+// chunks have landed, tile-k+1's 8 are still flying"; every wave waits
+// its own counter before the barrier, so after the barrier all waves'
+// tile-k chunks are visible.  Requires M%128==0, N%128==0, K%64==0,
+// K/64 >= 2 (caller-checked).
+__global__ __launch_bounds__(256) void k_gemm_nt_bf16_v4(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    ushort_t* __restrict__ C, int64_t M, int N, int K) {
+  __shared__ ushort_t lds[3 * 2 * BM * BK3];  // 3 buffers x (A,B)
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+  const int g_row_in_chunk = lane >> 3;
+  const int g_cb = (lane & 7) * 16;
+
+  const int n_tiles = N / BN;
+  const int64_t m_tiles = M / BM;
+  const int64_t total_tiles = m_tiles * n_tiles;
+  const int KT = K / BK3;
+
+#define GLDS4(buf, k0)                                                     \
+  do {                                                                     \
+    _Pragma("unroll") for (int c = 0; c < 4; ++c) {                        \
+      const int row = (wave * 4 + c) * 8 + g_row_in_chunk;                 \
+      const int scb = swz(row, g_cb);                                      \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              A + (tm + row) * K + (k0) + scb / 2),                        \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + (buf) * 2 * BM * BK3 + (wave * 4 + c) * 512),          \
+          16, 0, 0);                                                       \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              B + (int64_t)(tn + row) * K + (k0) + scb / 2),               \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + ((buf) * 2 + 1) * BM * BK3 + (wave * 4 + c) * 512),    \
+          16, 0, 0);                                                       \
+    }                                                                      \
+  } while (0)
+
+  for (int64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+    const int64_t tm = (tile / n_tiles) * BM;
+    const int tn = (int)(tile % n_tiles) * BN;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    GLDS4(0, 0);
+    if (KT > 1) GLDS4(1, BK3);
+
+    for (int kt = 0; kt < KT; ++kt) {
+      // wait for OWN tile-k chunks; leave tile-k+1's (if any) in flight
+      if (kt + 1 < KT)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+
+      const int buf = kt % 3;
+      const ushort_t* As_ = lds + buf * 2 * BM * BK3;
+      const ushort_t* Bs_ = lds + (buf * 2 + 1) * BM * BK3;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int ra = wm + i * 16 + frow;
+          const int rb = wn + i * 16 + frow;
+          const int cb = ks * 64 + fkb;
+          afrag[i] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(As_) + ra * 128 + swz(ra, cb));
+          bfrag[i] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(Bs_) + rb * 128 + swz(rb, cb));
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      }
+      // re-fill buffer (kt+2)%3 == (kt-1)%3: passing this iteration's
+      // barrier proves every wave finished reading tile k-1, so no
+      // second barrier is needed before overwriting its buffer
+      if (kt + 2 < KT) GLDS4((kt + 2) % 3, (int64_t)(kt + 2) * BK3);
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gc = tn + wn + j * 16 + (lane & 15);
+          C[gr * N + gc] = f2b(acc[i][j][r]);
+        }
+  }
+#undef GLDS4
+}
+
 inline int gemm_grid(int64_t M, int N) {
   int64_t tiles = ((M + BM - 1) / BM) * (int64_t)((N + BN - 1) / BN);
   if (tiles > 16384) tiles = 16384;
@@ -510,6 +624,12 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
 void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s) {
   hipLaunchKernelGGL(k_gemm_nt_bf16_v3, dim3(gemm_grid(M, N)), dim3(256), 0,
+                     s, A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s) {
+  hipLaunchKernelGGL(k_gemm_nt_bf16_v4, dim3(gemm_grid(M, N)), dim3(256), 0,
                      s, A, B, C, M, N, K);
 }
 

@@ -177,3 +177,34 @@ def test_gemm_v3_throughput_readout():
     t3 = t(lambda: ext().gemm_nt_bf16_v3(A, B, C))
     print(f"\n[gemm v2 vs v3-glds] {flops / t2 / 1e12:.1f} TF vs "
           f"{flops / t3 / 1e12:.1f} TF")
+
+
+def test_gemm_nt_v4_matches_v1_and_throughput():
+    M, N, K = 25088, 512, 512
+    torch.manual_seed(4)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C4 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v4(A, B, C4)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C4), (
+        (C1.float() - C4.float()).abs().max().item()
+    )
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t3 = t(lambda: ext().gemm_nt_bf16_v3(A, B, C4))
+    t4 = t(lambda: ext().gemm_nt_bf16_v4(A, B, C4))
+    print(f"\n[gemm v3 vs v4-span] {flops / t3 / 1e12:.1f} TF vs "
+          f"{flops / t4 / 1e12:.1f} TF")
