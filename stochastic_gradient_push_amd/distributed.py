@@ -474,8 +474,10 @@ class GossipDataParallel(Module):
 
         if self.gossip_flag.is_set():
             # interrupted gossip — re-arm and retry next iteration
-            # (reference distributed.py:359-364)
-            if self.gossip_ps_weight[0] == -1:
+            # (reference distributed.py:359-364).  The host-side flag
+            # avoids a device->host sync on the per-step hot path; the
+            # tensor sentinel stays authoritative for checkpoints.
+            if self.dist_config.pop("gossip_failed", False):
                 self.gossip_flag.clear()
                 self.params_mixed = True
                 self.gossiping = False
@@ -635,6 +637,7 @@ class GossipDataParallel(Module):
                 logger.warning(f"received runtime error {e}")
                 gossiper.clean_msg_buffers_()
                 gossip_ps_weight.fill_(-1)
+                dist_config["gossip_failed"] = True
             finally:
                 if gossip_stream is not None:
                     gossip_stream.synchronize()
