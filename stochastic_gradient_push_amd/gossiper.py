@@ -188,6 +188,14 @@ class Gossiper:
         self.refresh_mixing_weights_(residual)
         self.ps_weight = ps_weight
 
+        # training fast path (regular + uniform + residual): the
+        # residual-adjusted weight is exactly 1 and no ps-weight scalar
+        # travels, so the caller's buffer is sent as-is — zero-copy.
+        # Safe because the wrapper's flag protocol guarantees the buffer
+        # is not mutated until the exchange completes.
+        if residual and self.regular and self._mixing_manager.is_uniform():
+            return out_msg
+
         n = out_msg.numel()
         if not self.regular:
             assert self._send_buffer.numel() == n + 1
