@@ -4,14 +4,14 @@ Behavioral parity with reference gossip/distributed.py:39-589, rebuilt
 MI355X-first:
 
 * **One process per GPU** is the native mode (`torch.distributed` over
-  RCCL/xGMI).  The reference's single-process multi-GPU replica tier
-  (DataParallel-style replicate/scatter/gather, reference
-  distributed.py:91-99, 231-254) is intentionally not reproduced — it is
-  a PyTorch-1.x idiom superseded by process-per-GPU; the multi-process
-  intra-node tier (``nprocs_per_node > 1``: params broadcast from the
-  local master, grads all-reduced on a per-node group, gossip done by the
-  local master only, reference distributed.py:62-78, 278-296, 551-562) IS
-  supported.
+  RCCL/xGMI).  Both of the reference's intra-node tiers are supported
+  for completeness: the single-process multi-GPU replica tier
+  (``device_ids=[...]``, DataParallel-style scatter/parallel_apply/
+  gather with flat-buffer replica sync and grad reduction, reference
+  distributed.py:91-99, 231-254, 524-549) and the multi-process tier
+  (``nprocs_per_node > 1``: params broadcast from the local master,
+  grads all-reduced on a per-node group, gossip done by the local
+  master only, reference distributed.py:62-78, 278-296, 551-562).
 * **Flat parameter buffer**: all trainable params live as views of one
   contiguous device buffer (:class:`~..ops.flat.FlatParams`), so every
   push-sum state transition (bias, de-bias, residual-add, pack) is a
