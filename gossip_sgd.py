@@ -270,6 +270,7 @@ def make_dataloader(args, train=True):
             dataset, batch_size=args.batch_size, shuffle=False,
             num_workers=args.num_dataloader_workers,
             pin_memory=(args.device == "cuda"), sampler=sampler,
+            drop_last=True,  # fixed step shape (graph capture, lr scaling)
         )
         return loader, sampler
     return torch.utils.data.DataLoader(
@@ -393,6 +394,19 @@ class GraphedStep:
         return out.detach(), loss.detach()
 
     def __call__(self, batch, kl_target):
+        if self.sx is not None and batch.shape != self.sx.shape:
+            # ragged batch (shouldn't happen with drop_last, but never
+            # crash on it): run this one eagerly outside the graph
+            with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                out = self.inner(batch)
+            loss = self.criterion(out, kl_target)
+            loss.backward()
+            self.optimizer.step()
+            self.optimizer.zero_grad()
+            if self.gdp.distributed:
+                self.gdp._query_gossip_queue(non_blocking=self.gdp.asynch)
+                self.gdp.transfer_params()
+            return out.detach(), loss.detach()
         if self.sx is None:
             self.sx = batch.clone()
             self.st = kl_target.clone()
