@@ -72,3 +72,19 @@ def test_is_power_of():
     assert not is_power_of(12, 2)
     assert is_power_of(1, 0)
     assert not is_power_of(5, 1)
+
+
+def test_transformer_log_parser(tmp_path):
+    from visualization.plotting import parse_transformer_out
+
+    log = tmp_path / "train.log"
+    log.write_text(
+        "| epoch 001 | valid on 'valid' subset | nll_loss 9.123 "
+        "| num_updates 100\n"
+        "| epoch 002 | valid on 'valid' subset | nll_loss 7.5 "
+        "| num_updates 200\n"
+        "| training line without valid keyword | loss 3\n"
+    )
+    df = parse_transformer_out(str(log))
+    assert list(df["num_updates"]) == [100, 200]
+    assert df["valid_nll_loss"].iloc[1] == 7.5

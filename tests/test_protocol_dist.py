@@ -181,3 +181,55 @@ def _runtime_ppi_change(rank, world_size):
 
 def test_runtime_peers_per_itr_change():
     run_dist(_runtime_ppi_change, world_size=4)
+
+
+def _chaos_training(rank, world_size):
+    """Stress: nonzero-lr training while gossip randomly fails; training
+    must stay finite and keep making progress (sentinel retry path under
+    load)."""
+    import random
+
+    from stochastic_gradient_push_amd import GossipDataParallel
+
+    torch.manual_seed(rank)
+    gdp = GossipDataParallel(tiny_model(rank), push_sum=True)
+    opt = torch.optim.SGD(gdp.parameters(), lr=0.02)
+
+    gossiper = list(gdp.dist_config["gossipers"].values())[0]
+    original_mix = gossiper.mix
+    rng = random.Random(1234)  # same schedule on every rank
+
+    def flaky_mix(*a, **kw):
+        if rng.random() < 0.3:
+            raise RuntimeError("chaos")
+        return original_mix(*a, **kw)
+
+    with gdp.gossip_lock:
+        gossiper.mix = flaky_mix
+
+    torch.manual_seed(99)
+    w_true = torch.randn(6, 4)
+    torch.manual_seed(200 + rank)
+    x = torch.randn(16, 6)
+    y = x @ w_true
+
+    gdp.train()
+    first = last = None
+    for _ in range(25):
+        loss = ((gdp(x) - y) ** 2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        gdp.transfer_params()
+        if first is None:
+            first = loss.item()
+        last = loss.item()
+    gdp.sync_comms()
+    gdp.unbias()
+    assert torch.isfinite(gdp.flatp.flat).all()
+    assert last < first  # still optimizing through the failures
+    gdp.shutdown()
+
+
+def test_chaos_gossip_failures():
+    run_dist(_chaos_training, world_size=2)
