@@ -1,6 +1,7 @@
 """flatten/unflatten round-trips, Meter stats, is_power_of."""
 
 import math
+import os
 
 import pytest
 import torch
@@ -88,3 +89,34 @@ def test_transformer_log_parser(tmp_path):
     df = parse_transformer_out(str(log))
     assert list(df["num_updates"]) == [100, 200]
     assert df["valid_nll_loss"].iloc[1] == 7.5
+
+
+def test_plotting_renders_png(tmp_path):
+    """plot_itrs/plot_scaling produce image files from a real run's CSV."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update(RANK="0", WORLD_SIZE="1", MASTER_ADDR="127.0.0.1")
+    subprocess.run(
+        [
+            sys.executable, os.path.join(repo, "gossip_sgd.py"),
+            "--num_epochs", "1", "--num_iterations_per_training_epoch", "4",
+            "--batch_size", "2", "--synthetic_size", "16",
+            "--model", "resnet18", "--num_classes", "10",
+            "--image_size", "32", "--num_dataloader_workers", "0",
+            "--device", "cpu", "--checkpoint_dir", f"{tmp_path}/ck/",
+            "--num_itr_ignore", "0", "--graph_type", "-1",
+            "--train_fast", "True", "--print_freq", "1",
+            "--master_port", "29821",
+        ],
+        env=env, check=True, timeout=240, cwd=str(tmp_path),
+    )
+    from visualization.plotting import load_experiment, plot_itrs, plot_scaling
+
+    runs = load_experiment(f"{tmp_path}/ck/")
+    out = plot_itrs(runs, f"{tmp_path}/loss.png")
+    assert os.path.exists(out)
+    out2 = plot_scaling({1: 0.5, 2: 0.3, 4: 0.2}, f"{tmp_path}/scale.png")
+    assert os.path.exists(out2)
