@@ -49,7 +49,7 @@ def parse_args():
     )
     p.add_argument(
         "--conv-impl", type=str, default="miopen",
-        choices=["miopen", "gemm"],
+        choices=["miopen", "gemm", "mfma"],
         help="1x1 convs: MIOpen conv kernels vs hipBLASLt GEMM dispatch",
     )
     p.add_argument(

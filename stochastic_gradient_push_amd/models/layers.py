@@ -82,6 +82,80 @@ class GemmConv1x1(nn.Conv2d):
         )  # channels_last strides, zero-copy
 
 
+class _MfmaConv1x1Fn(torch.autograd.Function):
+    """1x1 conv through the hand-written MFMA NT GEMM
+    (ops/csrc/gemm1x1_kernels.hip, hardware-probe-verified):
+      fwd   y[M,Co] = x[M,Ci] @ W[Co,Ci]^T          (NT kernel)
+      dgrad dx[M,Ci] = dy[M,Co] @ W = dy @ (W^T)^T  (NT kernel, W^T)
+      wgrad dW[Co,Ci] = dy^T @ x                    (library GEMM — a
+            TN reduction over the huge M dim; hand-written TN kernel is
+            round-2 work, see docs/ROADMAP.md)
+    """
+
+    @staticmethod
+    def forward(ctx, x2d, weight):
+        from .. import ops as _o
+
+        k = _o._ext_for(x2d)
+        w_bf16 = weight.detach().to(torch.bfloat16)
+        y = torch.empty(
+            x2d.shape[0], weight.shape[0], device=x2d.device,
+            dtype=torch.bfloat16,
+        )
+        if weight.shape[1] % 64 == 0:
+            k.gemm_nt_bf16_v3(x2d, w_bf16, y)
+        else:
+            k.gemm_nt_bf16_v2(x2d, w_bf16, y)
+        ctx.save_for_backward(x2d, w_bf16)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import ops as _o
+
+        x2d, w_bf16 = ctx.saved_tensors
+        k = _o._ext_for(x2d)
+        dy = dy.contiguous()
+        dx = torch.empty_like(x2d)
+        wt = w_bf16.t().contiguous()  # [Ci, Co]
+        if wt.shape[1] % 64 == 0:
+            k.gemm_nt_bf16_v3(dy, wt, dx)
+        else:
+            k.gemm_nt_bf16_v2(dy, wt, dx)
+        dw = torch.matmul(dy.t(), x2d).float()
+        return dx, dw
+
+
+class MfmaConv1x1(nn.Conv2d):
+    """1x1 convolution on the hand-written MFMA GEMM kernels (north-star
+    conv path; behind ``conv_impl='mfma'`` — MIOpen remains the default
+    while the kernels trail hipBLASLt, see profiles/r01_summary.md)."""
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__(cin, cout, 1, stride=stride, bias=False)
+
+    def forward(self, x):
+        if not (
+            x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and self.in_channels % 32 == 0
+        ):
+            return super().forward(x)
+        strided = self.stride[0] != 1
+        x_in = x
+        if strided:
+            x_in = x[:, :, :: self.stride[0], :: self.stride[1]].contiguous(
+                memory_format=torch.channels_last
+            )
+        n, c, h, w = x_in.shape
+        x2d = x_in.permute(0, 2, 3, 1).reshape(n * h * w, c)
+        y2d = _MfmaConv1x1Fn.apply(x2d.contiguous(), self.weight.view(
+            self.out_channels, c
+        ))
+        return y2d.view(n, h, w, self.out_channels).permute(0, 3, 1, 2)
+
+
 def _fused_supported(x: torch.Tensor, C: int) -> bool:
     """The hand-written kernels need bf16 NHWC with C = 8 * 2^k <= 2048
     (the reduce kernel's thread geometry; every ResNet width qualifies)."""

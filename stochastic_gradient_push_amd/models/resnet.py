@@ -28,6 +28,10 @@ def conv1x1(cin: int, cout: int, stride: int = 1,
         from .layers import GemmConv1x1
 
         return GemmConv1x1(cin, cout, stride=stride)
+    if impl == "mfma":
+        from .layers import MfmaConv1x1
+
+        return MfmaConv1x1(cin, cout, stride=stride)
     return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
 
 
