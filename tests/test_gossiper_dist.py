@@ -159,3 +159,65 @@ def test_pushsum_eight_process_emulation(graph_cls):
     """Multi-node emulation (SURVEY tier): 8 CPU processes, larger
     topology, distributed averaging converges."""
     run_dist(_pushsum_average, world_size=8, args=("PushSum", graph_cls))
+
+
+def _communicate_helper(rank, world_size):
+    """utils.communicate: flatten per dtype, run the op, write back
+    (reference helpers.py:73-88)."""
+    from stochastic_gradient_push_amd.utils import communicate
+
+    tensors = [
+        torch.full((3, 3), float(rank)),
+        torch.full((5,), float(rank)).double(),
+    ]
+    communicate(tensors, lambda tensor: dist.all_reduce(tensor))
+    expected = sum(range(world_size))
+    for t in tensors:
+        assert torch.allclose(t, torch.full_like(t, float(expected)))
+
+
+def test_communicate_helper():
+    run_dist(_communicate_helper, world_size=2)
+
+
+def _nonregular_wire_weight(rank, world_size):
+    """Non-regular graphs transmit the push-sum weight as a trailing
+    scalar (reference gossiper.py:83-85, 132): with explicit weight
+    tracking the de-biased estimates still converge to the average."""
+    from stochastic_gradient_push_amd import gossiper as G
+    from stochastic_gradient_push_amd.graphs import (
+        NPeerDynamicDirectedExponentialGraph,
+    )
+
+    class NonRegularGraph(NPeerDynamicDirectedExponentialGraph):
+        def is_regular_graph(self):
+            return False  # forces ps-weight on the wire
+
+    torch.manual_seed(rank)
+    graph = NonRegularGraph(rank, world_size)
+    gossiper = G.PushSum(
+        torch.zeros(N), graph=graph, device=torch.device("cpu"),
+        rank=rank, world_size=world_size,
+    )
+    assert not gossiper.regular
+    assert gossiper.in_msg_buffer.numel() == N + 1
+
+    x = torch.randn(N)
+    w = torch.ones(1)
+    target = x.clone()
+    dist.all_reduce(target)
+    target /= world_size
+
+    for _ in range(60):
+        x, w = gossiper.mix(x.clone(), w.clone(), residual=False)
+        x = x.clone()
+        w = w.clone().reshape(1)
+
+    est = x / w
+    assert torch.allclose(est, target, atol=1e-3), (
+        f"rank {rank}: max err {(est - target).abs().max()}"
+    )
+
+
+def test_nonregular_wire_weight():
+    run_dist(_nonregular_wire_weight, world_size=4)
