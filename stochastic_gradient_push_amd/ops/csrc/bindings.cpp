@@ -56,6 +56,9 @@ void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
+                            float* partials, float* dw, int64_t M, int Co,
+                            int Ci, int split, hipStream_t s);
 }
 
 namespace {
@@ -199,6 +202,30 @@ void gemm_nt_bf16_v4(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), current_stream(A));
+}
+
+void gemm_tn_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
+                        torch::Tensor partials, torch::Tensor dw,
+                        int64_t split) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous()
+              && dy.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+              && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.size(0) == dy.size(0), "M mismatch");
+  const int64_t Co = dy.size(1), Ci = x.size(1);
+  TORCH_CHECK(dw.is_cuda() && dw.is_contiguous()
+              && dw.scalar_type() == torch::kFloat32
+              && dw.numel() == Co * Ci, "dw must be fp32 [Co*Ci]");
+  TORCH_CHECK(partials.is_cuda() && partials.is_contiguous()
+              && partials.scalar_type() == torch::kFloat32
+              && partials.numel() == split * Co * Ci,
+              "partials must be fp32 [split*Co*Ci]");
+  TORCH_CHECK(dy.size(0) % 32 == 0, "M must be a multiple of 32");
+  sgp_gemm_tn_wgrad_bf16(
+      reinterpret_cast<const ushort_t*>(dy.data_ptr()),
+      reinterpret_cast<const ushort_t*>(x.data_ptr()),
+      partials.data_ptr<float>(), dw.data_ptr<float>(), dy.size(0),
+      (int)Co, (int)Ci, (int)split, current_stream(dy));
 }
 
 void gemm_nt_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
@@ -374,6 +401,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "global_load_lds + st_16x32 swizzle variant (K % 64 == 0)");
   m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
         "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
+  m.def("gemm_tn_wgrad_bf16", &gemm_tn_wgrad_bf16,
+        "EXPERIMENTAL: dW = dy^T @ x with split-M partials (round-2 "
+        "validation pending)");
   m.def("pack_mix_bf16_", &pack_mix_bf16_,
         "x *= a; out_bf16 = bf16(x) (wire-format pack)");
   m.def("add_scale_bf16_", &add_scale_bf16_,

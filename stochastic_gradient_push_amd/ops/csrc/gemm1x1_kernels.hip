@@ -599,6 +599,133 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v4(
 #undef GLDS4
 }
 
+// ------------------------------------------- wgrad TN (EXPERIMENTAL)
+// dW[Co,Ci] = sum_m dy[m,co] * x[m,ci] — the 1x1-conv weight gradient.
+// K = M is huge, so blocks split the M range and store fp32 partials
+// [split][128][128]; a second kernel reduces them (deterministic, no
+// atomics — same pattern as the BN reductions).  Staging transposes on
+// write into the v1 kernel's padded [c][m] LDS image, so the MFMA inner
+// loop is identical to the NT kernel's.  Compile-verified this round;
+// hardware validation is round-2 work (tests marked gpu_experimental).
+__global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
+    const ushort_t* __restrict__ dy,  // [M, Co]
+    const ushort_t* __restrict__ x,   // [M, Ci]
+    float* __restrict__ partials,     // [split][Co, Ci] (tile-major ok)
+    int64_t M, int Co, int Ci, int split) {
+  __shared__ ushort_t As[BM * LDS_STRIDE];  // [co][m] transposed image
+  __shared__ ushort_t Bs[BN * LDS_STRIDE];  // [ci][m]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;  // co offset of the wave
+  const int wn = (wave & 1) * 64;   // ci offset
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;
+
+  const int co_tiles = (Co + BM - 1) / BM;
+  const int ci_tiles = (Ci + BN - 1) / BN;
+  const int64_t tiles = (int64_t)co_tiles * ci_tiles * split;
+
+  for (int64_t t = blockIdx.x; t < tiles; t += gridDim.x) {
+    const int s = (int)(t % split);
+    const int64_t ct = t / split;
+    const int tco = (int)(ct / ci_tiles) * BM;
+    const int tci = (int)(ct % ci_tiles) * BN;
+
+    const int64_t m0 = (M * s) / split;
+    const int64_t m1 = (M * (s + 1)) / split;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int64_t k0 = m0; k0 < m1; k0 += BK) {
+      __syncthreads();
+      // stage BK=32 m-rows x 128 channels per operand, transposing into
+      // the padded [channel][m] image; 2 segments x 2 operands/thread
+#pragma unroll
+      for (int seg4 = 0; seg4 < 2; ++seg4) {
+        const int seg = tid + seg4 * 256;  // 0..511
+        const int mrow = seg >> 4;         // 32 m-rows
+        const int c8 = (seg & 15) * 8;     // 16 channel-groups of 8
+        const int64_t gm = k0 + mrow;
+        // dy -> As
+        {
+          ushort_t tmp[8];
+          const bool ok = gm < m1;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int gc = tco + c8 + j;
+            tmp[j] = (ok && gc < Co)
+                ? dy[gm * Co + gc] : (ushort_t)0;
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            As[(c8 + j) * LDS_STRIDE + mrow] = tmp[j];
+        }
+        // x -> Bs
+        {
+          ushort_t tmp[8];
+          const bool ok = gm < m1;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int gc = tci + c8 + j;
+            tmp[j] = (ok && gc < Ci)
+                ? x[gm * Ci + gc] : (ushort_t)0;
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            Bs[(c8 + j) * LDS_STRIDE + mrow] = tmp[j];
+        }
+      }
+      __syncthreads();
+
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            As + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
+        bfrag[i] = *reinterpret_cast<const bf16x8*>(
+            Bs + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+
+    float* out = partials + (int64_t)s * Co * Ci;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int gco = tco + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gci = tci + wn + j * 16 + (lane & 15);
+          if (gco < Co && gci < Ci)
+            out[(int64_t)gco * Ci + gci] = acc[i][j][r];
+        }
+  }
+}
+
+__global__ void k_gemm_tn_reduce(const float* __restrict__ partials,
+                                 float* __restrict__ dw, int64_t numel,
+                                 int split) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    float s = 0.f;
+    for (int k = 0; k < split; ++k) s += partials[(int64_t)k * numel + i];
+    dw[i] = s;
+  }
+}
+
 inline int gemm_grid(int64_t M, int N) {
   int64_t tiles = ((M + BM - 1) / BM) * (int64_t)((N + BN - 1) / BN);
   if (tiles > 16384) tiles = 16384;
@@ -631,6 +758,21 @@ void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s) {
   hipLaunchKernelGGL(k_gemm_nt_bf16_v4, dim3(gemm_grid(M, N)), dim3(256), 0,
                      s, A, B, C, M, N, K);
+}
+
+void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
+                            float* partials, float* dw, int64_t M, int Co,
+                            int Ci, int split, hipStream_t s) {
+  const int64_t tiles =
+      (int64_t)((Co + 127) / 128) * ((Ci + 127) / 128) * split;
+  int grid = (int)(tiles > 16384 ? 16384 : tiles);
+  hipLaunchKernelGGL(k_gemm_tn_partial_bf16, dim3(grid), dim3(256), 0, s,
+                     dy, x, partials, M, Co, Ci, split);
+  const int64_t numel = (int64_t)Co * Ci;
+  int rgrid = (int)(((numel + 255) / 256) > 8192 ? 8192
+                                                 : (numel + 255) / 256);
+  hipLaunchKernelGGL(k_gemm_tn_reduce, dim3(rgrid < 1 ? 1 : rgrid),
+                     dim3(256), 0, s, partials, dw, numel, split);
 }
 
 void sgp_gemm_nt_bf16_v2(const ushort_t* A, const ushort_t* B, ushort_t* C,
