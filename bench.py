@@ -212,7 +212,11 @@ def main():
     else:
         def step():
             loss = compute_step(model)
-            if gdp is not None and args.algorithm != "osgp":
+            # kick the exchange unless the wrapper's overlap hooks already
+            # did (osgp with overlap=True kicks in the forward-pre hook;
+            # a capture-failed osgp has overlap=False and degenerates to
+            # synchronous SGP here — gossip must still run)
+            if gdp is not None and not gdp.overlap:
                 gdp.transfer_params()
             return loss
 

@@ -616,6 +616,12 @@ class GossipDataParallel(Module):
             logger.debug("received train-flag")
             _t0 = _time.perf_counter()
             try:
+                if dist_config["cpu_comm"] and gossip_stream is not None:
+                    # CPU-comm with CUDA training: the D2H copy into
+                    # gossip_params was enqueued on gossip_stream by
+                    # transfer_params(); gloo reads the host buffer
+                    # directly, so order it behind the copy here
+                    gossip_stream.synchronize()
                 if gossip_stream is not None:
                     with torch.cuda.stream(gossip_stream):
                         ps_weight, ps_factor = (

@@ -27,14 +27,23 @@ class NativeBatchNorm2d(nn.BatchNorm2d):
             num_features, eps=eps, momentum=momentum, affine=affine,
             track_running_stats=True,
         )
-        # drop the counter buffer: constant momentum never needs it and it
-        # costs one tiny kernel per BN per step
-        self.num_batches_tracked = None
+        # the counter buffer stays registered (so checkpoints remain
+        # interchangeable with stock nn.BatchNorm2d under strict
+        # load_state_dict) but is never incremented: forward() calls
+        # torch.batch_norm directly with a constant momentum, skipping
+        # the per-BN long-add kernel the stock forward launches
 
     def forward(self, x):
+        momentum = self.momentum
+        if momentum is None:
+            # cumulative moving average — the one case that needs the
+            # counter (and it's updated host-side, no extra kernel)
+            if self.training:
+                self.num_batches_tracked += 1
+            momentum = 1.0 / float(self.num_batches_tracked.clamp(min=1))
         return torch.batch_norm(
             x, self.weight, self.bias, self.running_mean, self.running_var,
-            self.training, self.momentum, self.eps,
+            self.training, momentum, self.eps,
             False,  # cudnn/miopen disabled -> native kernels
         )
 
@@ -252,7 +261,8 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             track_running_stats=True,
         )
         self.relu = relu
-        self.num_batches_tracked = None
+        # counter kept (checkpoint interchange with nn.BatchNorm2d) but
+        # never incremented — see NativeBatchNorm2d
 
     def forward(self, x, residual=None):
         if _fused_supported(x, self.num_features) and (

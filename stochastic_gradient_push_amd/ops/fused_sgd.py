@@ -72,6 +72,12 @@ class FusedSGD:
         changes when the step itself is inside a captured hipGraph."""
         lr = float(self.param_groups[0]["lr"])
         if self._lr_dev is not None and lr != self._lr_host:
+            if torch.cuda.is_current_stream_capturing():
+                # never bake the fill_ into a captured hipGraph — a
+                # captured constant would silently freeze the schedule
+                # on every replay; the post-capture sync_lr() call (or
+                # the next eager step) lands the update instead
+                return
             self._lr_dev.fill_(lr)
             self._lr_host = lr
 
