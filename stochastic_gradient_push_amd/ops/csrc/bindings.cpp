@@ -56,6 +56,8 @@ void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
                             float* partials, float* dw, int64_t M, int Co,
                             int Ci, int split, hipStream_t s);
@@ -188,6 +190,15 @@ void gemm_nt_bf16_v3(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
   gemm_nt_check(A, B, C);
   TORCH_CHECK(A.size(1) % 64 == 0, "v3 requires K %% 64 == 0");
   sgp_gemm_nt_bf16_v3(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), current_stream(A));
+}
+
+void gemm_nt_bf16_v5(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(1) % 64 == 0, "v5 requires K %% 64 == 0");
+  sgp_gemm_nt_bf16_v5(reinterpret_cast<const ushort_t*>(A.data_ptr()),
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), current_stream(A));
@@ -399,6 +410,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "pipelined (register-staged double-buffer) variant");
   m.def("gemm_nt_bf16_v3", &gemm_nt_bf16_v3,
         "global_load_lds + st_16x32 swizzle variant (K % 64 == 0)");
+  m.def("gemm_nt_bf16_v5", &gemm_nt_bf16_v5,
+        "NT bf16 MFMA GEMM v5 (v3 + XCD-aware tile remap)");
   m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
         "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
   m.def("gemm_tn_wgrad_bf16", &gemm_tn_wgrad_bf16,

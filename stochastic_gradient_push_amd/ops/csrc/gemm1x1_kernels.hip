@@ -320,6 +320,21 @@ __device__ __forceinline__ int swz(int row, int cb) {
   return cb ^ (((row >> 2) & 1) << 5);
 }
 
+// XCD-aware block->tile remap.  MI355X dispatches blocks round-robin
+// over the 8 XCDs (xcd = blockIdx % 8), each with a private L2.  The
+// tile order is n-fastest, so consecutive TILES share the same A
+// m-rows; without remapping, consecutive BLOCKS land on different XCDs
+// and every XCD streams its own copy of A from HBM (A is re-read
+// n_tiles times -> the 128x128 kernel is HBM-bound at 512-wide shapes).
+// Mapping bid -> (bid%8)*(grid/8) + bid/8 gives each XCD a contiguous
+// slab of tiles, so the A rows a slab touches stay resident in that
+// XCD's L2.  Requires gridDim.x % 8 == 0 (launcher guarantees).
+__device__ __forceinline__ int64_t xcd_virtual_bid() {
+  const int g8 = gridDim.x >> 3;
+  return (int64_t)(blockIdx.x & 7) * g8 + (blockIdx.x >> 3);
+}
+
+template <bool XCDMAP>
 __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v3(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     ushort_t* __restrict__ C, int64_t M, int N, int K) {
@@ -346,7 +361,8 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v3(
   const int64_t total_tiles = m_tiles * n_tiles;
   const int KT = K / BK3;
 
-  for (int64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+  const int64_t bid0 = XCDMAP ? xcd_virtual_bid() : (int64_t)blockIdx.x;
+  for (int64_t tile = bid0; tile < total_tiles; tile += gridDim.x) {
     const int64_t tm = (tile / n_tiles) * BM;
     const int tn = (int)(tile % n_tiles) * BN;
     const bool full = (tm + BM <= M) && (tn + BN <= N);
@@ -750,8 +766,16 @@ void sgp_gemm_nt_bf16(const ushort_t* A, const ushort_t* B, ushort_t* C,
 
 void sgp_gemm_nt_bf16_v3(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s) {
-  hipLaunchKernelGGL(k_gemm_nt_bf16_v3, dim3(gemm_grid(M, N)), dim3(256), 0,
-                     s, A, B, C, M, N, K);
+  hipLaunchKernelGGL(k_gemm_nt_bf16_v3<false>, dim3(gemm_grid(M, N)),
+                     dim3(256), 0, s, A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, hipStream_t s) {
+  // grid must be a multiple of 8 for the XCD remap
+  int grid = (gemm_grid(M, N) + 7) & ~7;
+  hipLaunchKernelGGL(k_gemm_nt_bf16_v3<true>, dim3(grid), dim3(256), 0, s,
+                     A, B, C, M, N, K);
 }
 
 void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,

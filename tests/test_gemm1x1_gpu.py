@@ -179,6 +179,47 @@ def test_gemm_v3_throughput_readout():
           f"{flops / t3 / 1e12:.1f} TF")
 
 
+@pytest.mark.parametrize("shape", [(25088, 512, 512), (1568, 2048, 512),
+                                   (256, 64, 64), (130, 72, 64)])
+def test_gemm_nt_v5_matches_v1(shape):
+    """v5 = v3 + XCD-aware tile remap: numerics must be bitwise v1."""
+    M, N, K = shape
+    torch.manual_seed(5)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C5 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v5(A, B, C5)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C5), (
+        (C1.float() - C5.float()).abs().max().item()
+    )
+
+
+def test_gemm_v5_throughput_readout():
+    M, N, K = 25088, 512, 512
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t3 = t(lambda: ext().gemm_nt_bf16_v3(A, B, C))
+    t5 = t(lambda: ext().gemm_nt_bf16_v5(A, B, C))
+    print(f"\n[gemm v3 vs v5-xcd] {flops / t3 / 1e12:.1f} TF vs "
+          f"{flops / t5 / 1e12:.1f} TF")
+
+
 def test_gemm_nt_v4_matches_v1_and_throughput():
     M, N, K = 25088, 512, 512
     torch.manual_seed(4)
