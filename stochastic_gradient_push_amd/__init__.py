@@ -24,7 +24,7 @@ from .graphs import (
     NPeerDynamicDirectedExponentialGraph,
     RingGraph,
 )
-from .mixing import MixingManager, UniformMixing
+from .mixing import MixingManager, UniformMixing, WeightedMixing
 
 __version__ = "0.1.0"
 
@@ -44,4 +44,5 @@ __all__ = [
     "RingGraph",
     "MixingManager",
     "UniformMixing",
+    "WeightedMixing",
 ]

@@ -39,6 +39,17 @@ class RcclTransport:
     ) -> None:
         self._comm.exchange(send, dests, recvs, srcs, blocking)
 
+    def exchange_multi(
+        self,
+        sends: List[torch.Tensor],
+        dests: List[int],
+        recvs: List[torch.Tensor],
+        srcs: List[int],
+        blocking: bool = True,
+    ) -> None:
+        """Per-destination send buffers (non-uniform mixing)."""
+        self._comm.exchange_multi(sends, dests, recvs, srcs, blocking)
+
     def synchronize(self) -> None:
         self._comm.synchronize()
 

@@ -612,6 +612,13 @@ class GossipDataParallel(Module):
             train_flag.wait()
             if stop_flag.is_set():
                 logger.debug("gossip thread stopping")
+                # drain in-flight sends and drop any pending recv so no
+                # request outlives the thread
+                try:
+                    gossiper.clean_msg_buffers_()
+                    gossiper._pending_req = None
+                except Exception:
+                    pass
                 return
             logger.debug("received train-flag")
             _t0 = _time.perf_counter()

@@ -141,6 +141,10 @@ class Gossiper:
         # (gossiper.py:98) so grouped receives land concurrently.
         self._recv_pool: List[torch.Tensor] = [self.in_msg_buffer.clone()]
         self._send_buffer = self.in_msg_buffer.clone()
+        # per-out-edge send buffers, used only by non-uniform mixing
+        # (each edge carries a differently-weighted message, reference
+        # gossiper.py:125-147 generator semantics)
+        self._send_pool_: List[torch.Tensor] = []
         self._pending_req = None
 
     # -- properties --------------------------------------------------------
@@ -205,17 +209,30 @@ class Gossiper:
             assert self._send_buffer.numel() == n
             self._send_buffer.copy_(out_msg)
 
-        assert self._mixing_manager.is_uniform(), (
-            "non-uniform mixing requires per-edge weighted messages; only "
-            "uniform mixing is currently supported (as in the reference's "
-            "sole MixingManager implementation)"
-        )
+        if self._mixing_manager.is_uniform():
+            if not residual:
+                weight = self.mixing_weights["uniform"]
+                self._send_buffer.mul_(weight.to(self._send_buffer.dtype))
+            # residual-adjusted uniform weight is w/lo == 1.0 exactly (the
+            # sender pre-scaled by lo) — no multiply, and no host sync.
+            return self._send_buffer
+
+        # non-uniform mixing: one weighted copy per out-edge (the
+        # reference's mix_out_msg_ generator, gossiper.py:125-147).
+        # _send_buffer holds the unweighted (msg [+ps]) staging.
+        sends = self._send_buffers(len(self.out_edges))
+        npp = self._graph_manager.nprocs_per_node
+        for buf, e in zip(sends, self.out_edges):
+            w = self.mixing_weights[e.dest if e.dest in self.mixing_weights
+                                    else (e.dest // npp) * npp]
+            torch.mul(self._send_buffer, w.to(buf.dtype), out=buf)
         if not residual:
-            weight = self.mixing_weights["uniform"]
-            self._send_buffer.mul_(weight.to(self._send_buffer.dtype))
-        # residual-adjusted uniform weight is w/lo == 1.0 exactly (the
-        # sender pre-scaled by lo) — no multiply, and no host sync.
-        return self._send_buffer
+            # loopback share: scale the staging buffer by lo in place so
+            # _loopback_msg returns the self-contribution
+            self._send_buffer.mul_(
+                self.mixing_weights["lo"].to(self._send_buffer.dtype)
+            )
+        return sends
 
     def _loopback_msg(self, residual: bool) -> Optional[torch.Tensor]:
         """Self-contribution ``lo * out_msg`` (reference gossiper.py:135-136);
@@ -230,6 +247,11 @@ class Gossiper:
         while len(self._recv_pool) < n:
             self._recv_pool.append(self.in_msg_buffer.clone())
         return self._recv_pool[:n]
+
+    def _send_buffers(self, n: int) -> List[torch.Tensor]:
+        while len(self._send_pool_) < n:
+            self._send_pool_.append(self.in_msg_buffer.clone())
+        return self._send_pool_[:n]
 
     # -- cleanup / parse ----------------------------------------------------
 
@@ -284,21 +306,23 @@ class Gossiper:
                 self.in_msg_buffer.zero_()
             recvs = self._recv_buffers(len(self.in_edges))
 
+        per_edge = isinstance(send_msg, list)
         if self.transport is not None:
             for e in self.out_edges:
                 assert e.src == self.rank
-            self.transport.exchange(
-                send_msg,
-                [e.dest for e in self.out_edges],
-                recvs,
-                [e.src for e in self.in_edges],
-            )
+            dests = [e.dest for e in self.out_edges]
+            srcs = [e.src for e in self.in_edges]
+            if per_edge:
+                self.transport.exchange_multi(send_msg, dests, recvs, srcs)
+            else:
+                self.transport.exchange(send_msg, dests, recvs, srcs)
         else:
             ops = []
-            for e in self.out_edges:
+            for i, e in enumerate(self.out_edges):
                 assert e.src == self.rank
+                buf = send_msg[i] if per_edge else send_msg
                 ops.append(
-                    dist.P2POp(dist.isend, send_msg, e.dest, group=self.group)
+                    dist.P2POp(dist.isend, buf, e.dest, group=self.group)
                 )
             for buf, e in zip(recvs, self.in_edges):
                 ops.append(
@@ -331,6 +355,11 @@ class PushSum(Gossiper):
         send = self._prep_out_msg(out_msg, ps_weight, residual)
         self._exchange(send, residual)
         self.refresh_peers_()
+        # re-derive weights for the NEW peer set so mixing_weights['lo']
+        # (read by the wrapper as next round's pre-scale factor) matches
+        # the lo the next mix will divide by — load-bearing for
+        # non-uniform mixing on dynamic graphs, where lo varies per set
+        self.refresh_mixing_weights_(residual)
         self.clean_msg_buffers_()
         return self.parse_in_msg_buffer(residual)
 
@@ -374,6 +403,8 @@ class BilatPushPull(Gossiper):
 
         if not self.passive:
             send = self._prep_out_msg(out_msg, 1.0, residual=True)
+            if isinstance(send, list):
+                send = send[0]  # single out-edge by contract
             ops = [
                 dist.P2POp(dist.isend, send, out_edge.dest, group=self.group),
                 dist.P2POp(
@@ -391,6 +422,8 @@ class BilatPushPull(Gossiper):
                 )
             if self._pending_req.is_completed():
                 send = self._prep_out_msg(out_msg, 1.0, residual=True)
+                if isinstance(send, list):
+                    send = send[0]
                 if self.logger is not None:
                     self.logger.debug(f"req. completed; sending to {out_edge}")
                 dist.send(tensor=send, dst=out_edge.dest, group=self.group)

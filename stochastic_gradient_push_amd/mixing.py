@@ -59,3 +59,56 @@ class UniformMixing(MixingManager):
 
     def is_uniform(self) -> bool:
         return True
+
+
+class WeightedMixing(MixingManager):
+    """Non-uniform column-stochastic mixing with explicit per-peer weights.
+
+    The reference's ``MixingManager`` is an extension point whose message
+    machinery (``mix_out_msg_``, reference gossiper.py:125-147) supports
+    per-edge weights even though only ``UniformMixing`` ships; this class
+    restores that capability here.
+
+    :param weights: dict mapping each possible out-peer NODE rank to its
+        mixing weight.  Column-stochasticity must hold per iteration over
+        the graph's *currently active* out-peers: the self-weight is
+        ``lo = 1 - sum(weights[p] for active p)`` and must stay positive
+        for every peer set the graph can activate.
+
+    Non-uniform mixing is never "regular", so the push-sum weight always
+    travels on the wire (scaled per edge, like the message itself).
+    """
+
+    def __init__(self, graph: GraphManager, device, weights: Dict[int, float]):
+        super().__init__(graph, device)
+        if any(w <= 0 for w in weights.values()):
+            raise ValueError("all mixing weights must be positive")
+        self._peer_weights = {int(k): float(v) for k, v in weights.items()}
+
+    def is_uniform(self) -> bool:
+        return False
+
+    def get_mixing_weights(
+        self, residual_adjusted: bool = True
+    ) -> Dict[object, torch.Tensor]:
+        out_peers, _ = self.graph_manager.get_peers()
+        missing = [op for op in out_peers if op not in self._peer_weights]
+        if missing:
+            raise KeyError(
+                f"no mixing weight for current out-peer(s) {missing}; "
+                "WeightedMixing needs a weight for every peer the graph "
+                "can activate"
+            )
+        total = sum(self._peer_weights[op] for op in out_peers)
+        if not total < 1.0 - 1e-9:
+            raise ValueError(
+                f"active out-peer weights sum to {total}; must be < 1 so "
+                "the self-weight stays positive"
+            )
+        lo = torch.tensor([1.0 - total], device=self.device)
+        weights: Dict[object, torch.Tensor] = {"lo": lo.clone()}
+        npp = self.graph_manager.nprocs_per_node
+        for op in out_peers:
+            w = torch.tensor([self._peer_weights[op]], device=self.device)
+            weights[op * npp] = (w / lo) if residual_adjusted else w
+        return weights

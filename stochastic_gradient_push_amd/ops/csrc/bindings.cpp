@@ -58,6 +58,8 @@ void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
+void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, int span, hipStream_t s);
 void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
                             float* partials, float* dw, int64_t M, int Co,
                             int Ci, int split, hipStream_t s);
@@ -202,6 +204,19 @@ void gemm_nt_bf16_v5(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), current_stream(A));
+}
+
+void gemm_nt_bf16_v6(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                     bool span) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(0) % 256 == 0 && B.size(0) % 128 == 0
+              && A.size(1) % 64 == 0,
+              "v6 requires M %% 256 == 0, N %% 128 == 0, K %% 64 == 0");
+  sgp_gemm_nt_bf16_v6(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), span ? 1 : 0,
+                      current_stream(A));
 }
 
 void gemm_nt_bf16_v4(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
@@ -412,6 +427,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "global_load_lds + st_16x32 swizzle variant (K % 64 == 0)");
   m.def("gemm_nt_bf16_v5", &gemm_nt_bf16_v5,
         "NT bf16 MFMA GEMM v5 (v3 + XCD-aware tile remap)");
+  m.def("gemm_nt_bf16_v6", &gemm_nt_bf16_v6, py::arg("A"), py::arg("B"),
+        py::arg("C"), py::arg("span") = false,
+        "NT bf16 MFMA GEMM v6 (256x128 tile, 8 waves; span=3-buf "
+        "barrier-crossing glds)");
   m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
         "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
   m.def("gemm_tn_wgrad_bf16", &gemm_tn_wgrad_bf16,

@@ -136,6 +136,51 @@ class RcclComm {
     }
   }
 
+  // Per-destination send buffers (non-uniform mixing: each out-edge
+  // carries a differently-weighted message).  Same grouping/stream
+  // semantics as exchange().
+  void exchange_multi(std::vector<torch::Tensor> sends,
+                      std::vector<int64_t> dests,
+                      std::vector<torch::Tensor> recvs,
+                      std::vector<int64_t> srcs, bool blocking = true) {
+    TORCH_CHECK(sends.size() == dests.size(), "one send per dest");
+    TORCH_CHECK(recvs.size() == srcs.size());
+    TORCH_CHECK(!sends.empty() || !recvs.empty());
+    const torch::Tensor& proto = sends.empty() ? recvs[0] : sends[0];
+    const ncclDataType_t dt = nccl_dtype(proto);
+    const size_t n = (size_t)proto.numel();
+
+    hipStream_t cur =
+        c10::hip::getCurrentHIPStream(proto.device().index()).stream();
+    HIP_CHECK(hipEventRecord(ready_ev_, cur));
+    HIP_CHECK(hipStreamWaitEvent(stream_, ready_ev_, 0));
+
+    {
+      py::gil_scoped_release nogil;
+      ncclGroupStart();
+      for (size_t i = 0; i < sends.size(); ++i) {
+        TORCH_CHECK(sends[i].is_cuda() && sends[i].is_contiguous());
+        TORCH_CHECK((size_t)sends[i].numel() == n, "send size mismatch");
+        NCCL_CHECK(ncclSend(sends[i].data_ptr(), n, dt, (int)dests[i],
+                            comm_, stream_));
+      }
+      for (size_t i = 0; i < recvs.size(); ++i) {
+        TORCH_CHECK(recvs[i].is_cuda() && recvs[i].is_contiguous());
+        TORCH_CHECK((size_t)recvs[i].numel() == n, "recv size mismatch");
+        NCCL_CHECK(ncclRecv(recvs[i].data_ptr(), n, dt, (int)srcs[i],
+                            comm_, stream_));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+      if (blocking) {
+        HIP_CHECK(hipStreamSynchronize(stream_));
+      }
+    }
+    if (!blocking) {
+      HIP_CHECK(hipEventRecord(done_ev_, stream_));
+      HIP_CHECK(hipStreamWaitEvent(cur, done_ev_, 0));
+    }
+  }
+
   void synchronize() {
     py::gil_scoped_release nogil;
     HIP_CHECK(hipStreamSynchronize(stream_));
@@ -169,6 +214,9 @@ void init_comm_core(py::module_& m) {
            py::arg("unique_id"), py::arg("rank"), py::arg("world_size"),
            py::arg("device"))
       .def("exchange", &RcclComm::exchange, py::arg("send"),
+           py::arg("dests"), py::arg("recvs"), py::arg("srcs"),
+           py::arg("blocking") = true)
+      .def("exchange_multi", &RcclComm::exchange_multi, py::arg("sends"),
            py::arg("dests"), py::arg("recvs"), py::arg("srcs"),
            py::arg("blocking") = true)
       .def("synchronize", &RcclComm::synchronize)

@@ -615,6 +615,148 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v4(
 #undef GLDS4
 }
 
+// ------------------------------------- gemm v6 (256x128 tile, 8 waves)
+// The 128x128 tile's LDS-staging traffic (2 x 16 KB per 64-K-step for
+// 4.2 MFLOP -> 66 flops/byte) caps it below hipBLASLt.  256x128 with
+// 512 threads stages 48 KB per step for 2x the flops (87 flops/byte),
+// keeps the same per-wave 64x64 accumulator geometry (8 waves in 4x2),
+// and stays at 2 waves/SIMD.  One workgroup per CU.
+//   SPAN=false: 2 LDS buffers (96 KB), __syncthreads pipeline (drains
+//               glds via its vmcnt fence).
+//   SPAN=true:  3 buffers (144 KB of the 160 KB LDS), raw s_barrier +
+//               counted vmcnt keeps one tile of glds in flight across
+//               the barrier — the guide's 1-block/CU-regime lever (v4
+//               showed it loses at 2 blocks/CU; here it can win).
+// Full tiles only: caller routes M%256 || N%128 || K%64 != 0 to v5.
+#define BM6 256
+#define BN6 128
+#define TILE6 (BM6 + BN6)          // 384 rows of 128 B = 48 KB / buffer
+
+template <bool SPAN>
+__global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    ushort_t* __restrict__ C, int64_t M, int N, int K) {
+  __shared__ ushort_t lds[(SPAN ? 3 : 2) * TILE6 * BK3];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;         // 8 waves in 4x2
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;   // 0,64,128,192
+  const int wn = (wave & 1) * 64;    // 0,64
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+  const int g_row_in_chunk = lane >> 3;
+  const int g_cb = (lane & 7) * 16;
+
+  const int n_tiles = N / BN6;
+  const int64_t m_tiles = M / BM6;
+  const int64_t total_tiles = m_tiles * n_tiles;
+  const int KT = K / BK3;
+
+  // per-wave glds assignment: wave w stages A chunks 4w..4w+3 and
+  // B chunks 2w..2w+1 (chunk = 8 rows x 128 B = 1 KB); 6 glds per wave
+  // per tile step.
+#define GLDS6(buf, k0)                                                     \
+  do {                                                                     \
+    _Pragma("unroll") for (int c = 0; c < 4; ++c) {                        \
+      const int row = (wave * 4 + c) * 8 + g_row_in_chunk;                 \
+      const int scb = swz(row, g_cb);                                      \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              A + (tm + row) * K + (k0) + scb / 2),                        \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + (buf) * TILE6 * BK3 + (wave * 4 + c) * 512),           \
+          16, 0, 0);                                                       \
+    }                                                                      \
+    _Pragma("unroll") for (int c = 0; c < 2; ++c) {                        \
+      const int row = (wave * 2 + c) * 8 + g_row_in_chunk;                 \
+      const int scb = swz(row, g_cb);                                      \
+      __builtin_amdgcn_global_load_lds(                                    \
+          (const __attribute__((address_space(1))) unsigned int*)(         \
+              B + (int64_t)(tn + row) * K + (k0) + scb / 2),               \
+          (__attribute__((address_space(3))) unsigned int*)(               \
+              lds + (buf) * TILE6 * BK3 + BM6 * BK3                        \
+              + (wave * 2 + c) * 512),                                     \
+          16, 0, 0);                                                       \
+    }                                                                      \
+  } while (0)
+
+  const int64_t bid0 = xcd_virtual_bid();
+  for (int64_t tile = bid0; tile < total_tiles; tile += gridDim.x) {
+    const int64_t tm = (tile / n_tiles) * BM6;
+    const int tn = (int)(tile % n_tiles) * BN6;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    if (SPAN) {
+      GLDS6(0, 0);
+      if (KT > 1) GLDS6(1, BK3);
+    } else {
+      GLDS6(0, 0);
+    }
+
+    for (int kt = 0; kt < KT; ++kt) {
+      if (SPAN) {
+        if (kt + 1 < KT)
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      } else {
+        __syncthreads();  // drains in-flight glds (vmcnt fence)
+        if (kt + 1 < KT) GLDS6((kt & 1) ^ 1, (int64_t)(kt + 1) * BK3);
+      }
+
+      const int buf = SPAN ? (kt % 3) : (kt & 1);
+      const ushort_t* As_ = lds + buf * TILE6 * BK3;
+      const ushort_t* Bs_ = As_ + BM6 * BK3;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int ra = wm + i * 16 + frow;
+          const int rb = wn + i * 16 + frow;
+          const int cb = ks * 64 + fkb;
+          afrag[i] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(As_) + ra * 128 + swz(ra, cb));
+          bfrag[i] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(Bs_) + rb * 128 + swz(rb, cb));
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      }
+      if (SPAN) {
+        // passing this round's barrier proves all waves read tile kt-1:
+        // its buffer (kt+2)%3 is safe to refill with no extra barrier
+        if (kt + 2 < KT) GLDS6((kt + 2) % 3, (int64_t)(kt + 2) * BK3);
+      }
+      // non-SPAN: the next iteration's __syncthreads() orders the
+      // buffer swap (v3 structure — no trailing barrier needed)
+    }
+#undef GLDS6
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gc = tn + wn + j * 16 + (lane & 15);
+          C[gr * N + gc] = f2b(acc[i][j][r]);
+        }
+  }
+}
+
 // ------------------------------------------- wgrad TN (EXPERIMENTAL)
 // dW[Co,Ci] = sum_m dy[m,co] * x[m,ci] — the 1x1-conv weight gradient.
 // K = M is huge, so blocks split the M range and store fp32 partials
@@ -776,6 +918,20 @@ void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
   int grid = (gemm_grid(M, N) + 7) & ~7;
   hipLaunchKernelGGL(k_gemm_nt_bf16_v3<true>, dim3(grid), dim3(256), 0, s,
                      A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, int span, hipStream_t s) {
+  int64_t tiles = (M / BM6) * (int64_t)(N / BN6);
+  if (tiles > 16384) tiles = 16384;
+  int grid = (int)((tiles + 7) & ~7);
+  if (grid < 8) grid = 8;
+  if (span)
+    hipLaunchKernelGGL(k_gemm_nt_bf16_v6<true>, dim3(grid), dim3(512), 0, s,
+                       A, B, C, M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_nt_bf16_v6<false>, dim3(grid), dim3(512), 0, s,
+                       A, B, C, M, N, K);
 }
 
 void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,

@@ -220,6 +220,50 @@ def test_gemm_v5_throughput_readout():
           f"{flops / t5 / 1e12:.1f} TF")
 
 
+@pytest.mark.parametrize("span", [False, True])
+@pytest.mark.parametrize("shape", [(25088, 512, 512), (6272, 1024, 256),
+                                   (100352, 256, 64), (512, 128, 64)])
+def test_gemm_nt_v6_matches_v1(shape, span):
+    """v6 (256x128 tile, 8 waves; optional barrier-span glds) must be
+    bitwise v1 on full-tile shapes."""
+    M, N, K = shape
+    torch.manual_seed(6)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C6 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v6(A, B, C6, span=span)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C6), (
+        span, (C1.float() - C6.float()).abs().max().item()
+    )
+
+
+def test_gemm_v6_throughput_readout():
+    M, N, K = 25088, 512, 512
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t5 = t(lambda: ext().gemm_nt_bf16_v5(A, B, C))
+    t6 = t(lambda: ext().gemm_nt_bf16_v6(A, B, C, span=False))
+    t6s = t(lambda: ext().gemm_nt_bf16_v6(A, B, C, span=True))
+    print(f"\n[gemm v5 vs v6 vs v6-span] {flops / t5 / 1e12:.1f} vs "
+          f"{flops / t6 / 1e12:.1f} vs {flops / t6s / 1e12:.1f} TF")
+
+
 def test_gemm_nt_v4_matches_v1_and_throughput():
     M, N, K = 25088, 512, 512
     torch.manual_seed(4)

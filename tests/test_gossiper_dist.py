@@ -221,3 +221,102 @@ def _nonregular_wire_weight(rank, world_size):
 
 def test_nonregular_wire_weight():
     run_dist(_nonregular_wire_weight, world_size=4)
+
+
+def _weighted_average(rank, world_size):
+    """Non-uniform mixing (WeightedMixing): per-edge weighted messages,
+    ps-weight on the wire; de-biased estimates still converge to the
+    true average (push-sum is exact for any column-stochastic mixing)."""
+    from stochastic_gradient_push_amd import gossiper as G
+    from stochastic_gradient_push_amd.graphs import (
+        NPeerDynamicDirectedExponentialGraph,
+    )
+    from stochastic_gradient_push_amd.mixing import WeightedMixing
+
+    torch.manual_seed(20 + rank)
+    graph = NPeerDynamicDirectedExponentialGraph(rank, world_size)
+    # possible out-peers at distances 2^i; deliberately asymmetric weights
+    weights = {
+        (rank + 1) % world_size: 0.35,
+        (rank + 2) % world_size: 0.20,
+    }
+    mixing = WeightedMixing(graph, torch.device("cpu"), weights)
+    gossiper = G.PushSum(
+        torch.zeros(N), graph=graph, mixing=mixing,
+        device=torch.device("cpu"), rank=rank, world_size=world_size,
+    )
+    assert not gossiper.regular  # non-uniform => ps-weight on the wire
+
+    x = torch.randn(N)
+    w = torch.ones(1)
+    target = x.clone()
+    dist.all_reduce(target)
+    target /= world_size
+
+    for _ in range(80):
+        x, w = gossiper.mix(x.clone(), w.clone(), residual=False)
+        x = x.clone()
+        w = w.clone().reshape(1)
+
+    est = x / w
+    assert torch.allclose(est, target, atol=1e-3), (
+        f"rank {rank}: max err {(est - target).abs().max()}"
+    )
+
+
+def test_weighted_mixing_converges():
+    run_dist(_weighted_average, world_size=4)
+
+
+def _weighted_mass_conservation(rank, world_size):
+    """Residual protocol with non-uniform mixing: the global sums of x
+    and of the push-sum weights are conserved every iteration even
+    though lo varies with the rotating peer set."""
+    from stochastic_gradient_push_amd import gossiper as G
+    from stochastic_gradient_push_amd.graphs import (
+        NPeerDynamicDirectedExponentialGraph,
+    )
+    from stochastic_gradient_push_amd.mixing import WeightedMixing
+
+    torch.manual_seed(40 + rank)
+    graph = NPeerDynamicDirectedExponentialGraph(rank, world_size)
+    weights = {
+        (rank + 1) % world_size: 0.35,
+        (rank + 2) % world_size: 0.20,
+    }
+    mixing = WeightedMixing(graph, torch.device("cpu"), weights)
+    gossiper = G.PushSum(
+        torch.zeros(N), graph=graph, mixing=mixing,
+        device=torch.device("cpu"), rank=rank, world_size=world_size,
+    )
+
+    x = torch.randn(N)
+    ps = torch.ones(1)
+    total0 = x.clone()
+    dist.all_reduce(total0)
+    w_total0 = torch.tensor([float(world_size)])
+
+    for _ in range(12):
+        # explicit (non-lazy) residual protocol: pre-scale by THIS
+        # round's lo (exported by the gossiper for the current peer
+        # set), exchange, then merge the received residuals
+        lo = gossiper.mixing_weights["lo"].clone()
+        x = x * lo
+        ps = ps * lo
+        in_msg, w_recv = gossiper.mix(x.clone(), ps.clone(), residual=True)
+        x = x + in_msg
+        ps = (ps + w_recv).reshape(1)
+
+        xs = x.clone()
+        dist.all_reduce(xs)
+        ws = ps.clone()
+        dist.all_reduce(ws)
+        assert torch.allclose(xs, total0, atol=1e-4), f"rank {rank}"
+        assert torch.allclose(ws, w_total0, atol=1e-5), f"rank {rank}"
+
+    est = x / ps
+    assert torch.allclose(est, total0 / world_size, atol=0.1)
+
+
+def test_weighted_mixing_mass_conserved():
+    run_dist(_weighted_mass_conservation, world_size=4)

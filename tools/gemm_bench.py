@@ -7,11 +7,14 @@ torch.matmul (hipBLASLt), prints a TF table.  Run on the GPU box:
 """
 
 import argparse
+import os
+import sys
 import time
 
 import torch
 
-from stochastic_gradient_push_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from stochastic_gradient_push_amd import ops  # noqa: E402
 
 
 def timeit(f, n=30, warm=5):
@@ -54,27 +57,30 @@ def main():
         (25088, 512, 512),         # the ladder's reference shape
     ]
 
-    print(f"{'M':>8} {'N':>5} {'K':>5} | {'v1':>6} {'v2':>6} {'v3':>6} "
-          f"{'v5':>6} {'lib':>6}  (TFLOP/s)")
+    cols = ("v2", "v3", "v5", "v6", "v6s", "lib")
+    print(f"{'M':>8} {'N':>5} {'K':>5} | " + " ".join(
+        f"{c:>6}" for c in cols) + "  (TFLOP/s)")
     for M, N, K in shapes:
         A = torch.randn(M, K, device=dev).to(torch.bfloat16)
         B = torch.randn(N, K, device=dev).to(torch.bfloat16)
         C = torch.zeros(M, N, device=dev, dtype=torch.bfloat16)
         fl = 2.0 * M * N * K
-        r = {}
-        r["v1"] = fl / timeit(lambda: ext.gemm_nt_bf16(A, B, C)) / 1e12
+        r = dict.fromkeys(cols, 0.0)
         r["v2"] = fl / timeit(lambda: ext.gemm_nt_bf16_v2(A, B, C)) / 1e12
         if K % 64 == 0:
             r["v3"] = fl / timeit(
                 lambda: ext.gemm_nt_bf16_v3(A, B, C)) / 1e12
             r["v5"] = fl / timeit(
                 lambda: ext.gemm_nt_bf16_v5(A, B, C)) / 1e12
-        else:
-            r["v3"] = r["v5"] = 0.0
+            if M % 256 == 0 and N % 128 == 0:
+                r["v6"] = fl / timeit(lambda: ext.gemm_nt_bf16_v6(
+                    A, B, C, span=False)) / 1e12
+                r["v6s"] = fl / timeit(lambda: ext.gemm_nt_bf16_v6(
+                    A, B, C, span=True)) / 1e12
         Bt = B.t().contiguous().t()
         r["lib"] = fl / timeit(lambda: torch.matmul(A, Bt)) / 1e12
         print(f"{M:>8} {N:>5} {K:>5} | " + " ".join(
-            f"{r[k]:>6.0f}" for k in ("v1", "v2", "v3", "v5", "lib")))
+            f"{r[k]:>6.0f}" for k in cols))
 
     # wgrad TN: dW[Co,Ci] = dy^T @ x on the same shapes (Co=N, Ci=K)
     print(f"\nwgrad TN (split={args.wgrad_split}):")
