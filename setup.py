@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "comm_core.cpp"),
         os.path.join(CSRC, "gossip_kernels.hip"),
         os.path.join(CSRC, "gemm1x1_kernels.hip"),
+        os.path.join(CSRC, "conv3x3_kernels.hip"),
         os.path.join(CSRC, "bn_kernels.hip"),
     ],
     libraries=["rccl"],

@@ -92,29 +92,20 @@ class GemmConv1x1(nn.Conv2d):
 
 
 class _MfmaConv1x1Fn(torch.autograd.Function):
-    """1x1 conv through the hand-written MFMA NT GEMM
+    """1x1 conv entirely on the hand-written MFMA GEMMs
     (ops/csrc/gemm1x1_kernels.hip, hardware-probe-verified):
       fwd   y[M,Co] = x[M,Ci] @ W[Co,Ci]^T          (NT kernel)
       dgrad dx[M,Ci] = dy[M,Co] @ W = dy @ (W^T)^T  (NT kernel, W^T)
-      wgrad dW[Co,Ci] = dy^T @ x                    (library GEMM — a
-            TN reduction over the huge M dim; hand-written TN kernel is
-            round-2 work, see docs/ROADMAP.md)
+      wgrad dW[Co,Ci] = dy^T @ x                    (split-M TN kernel
+            with deterministic partial reduce, hardware-validated)
     """
 
     @staticmethod
     def forward(ctx, x2d, weight):
         from .. import ops as _o
 
-        k = _o._ext_for(x2d)
         w_bf16 = weight.detach().to(torch.bfloat16)
-        y = torch.empty(
-            x2d.shape[0], weight.shape[0], device=x2d.device,
-            dtype=torch.bfloat16,
-        )
-        if weight.shape[1] % 64 == 0:
-            k.gemm_nt_bf16_v3(x2d, w_bf16, y)
-        else:
-            k.gemm_nt_bf16_v2(x2d, w_bf16, y)
+        y = _o.gemm_nt(x2d, w_bf16)
         ctx.save_for_backward(x2d, w_bf16)
         return y
 
@@ -123,15 +114,10 @@ class _MfmaConv1x1Fn(torch.autograd.Function):
         from .. import ops as _o
 
         x2d, w_bf16 = ctx.saved_tensors
-        k = _o._ext_for(x2d)
         dy = dy.contiguous()
-        dx = torch.empty_like(x2d)
         wt = w_bf16.t().contiguous()  # [Ci, Co]
-        if wt.shape[1] % 64 == 0:
-            k.gemm_nt_bf16_v3(dy, wt, dx)
-        else:
-            k.gemm_nt_bf16_v2(dy, wt, dx)
-        dw = torch.matmul(dy.t(), x2d).float()
+        dx = _o.gemm_nt(dy, wt)
+        dw = _o.gemm_tn_wgrad(dy, x2d)
         return dx, dw
 
 
@@ -163,6 +149,84 @@ class MfmaConv1x1(nn.Conv2d):
             self.out_channels, c
         ))
         return y2d.view(n, h, w, self.out_channels).permute(0, 3, 1, 2)
+
+
+class _MfmaConv3x3Fn(torch.autograd.Function):
+    """3x3 pad-1 conv on the hand-written MFMA implicit-GEMM kernel
+    (ops/csrc/conv3x3_kernels.hip).
+
+    fwd:   implicit-GEMM kernel (stride 1 or 2)
+    dgrad: stride 1 -> the SAME kernel on dy with the rotated/transposed
+           weight w_rot[ci,r',s',co] = w[co,2-r',2-s',ci];
+           stride 2 -> torch.nn.grad.conv2d_input (MIOpen fallback)
+    wgrad: torch.nn.grad.conv2d_weight (MIOpen; implicit-TN kernel is
+           future work)
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        from .. import ops as _o
+
+        k = _o._ext_for(x)
+        n, ci, h, w = x.shape
+        co = weight.shape[0]
+        ho = (h - 1) // stride + 1
+        wo = (w - 1) // stride + 1
+        # [Co,3,3,Ci] contiguous = channels_last view of the weight
+        w_bf = (
+            weight.detach().permute(0, 2, 3, 1).contiguous()
+            .to(torch.bfloat16)
+        )
+        y = torch.empty(
+            n, co, ho, wo, device=x.device, dtype=torch.bfloat16,
+            memory_format=torch.channels_last,
+        )
+        k.conv3x3_nhwc_bf16(x, w_bf, y, stride)
+        ctx.save_for_backward(x, w_bf)
+        ctx.c3_stride = stride
+        ctx.c3_wdtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import ops as _o
+
+        x, w_bf = ctx.saved_tensors
+        stride = ctx.c3_stride
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        k = _o._ext_for(x)
+        w_nchw = w_bf.permute(0, 3, 1, 2)  # [Co,Ci,3,3] (strided view ok)
+        if stride == 1:
+            w_rot = w_bf.flip(1, 2).permute(3, 1, 2, 0).contiguous()
+            dx = torch.empty_like(x)
+            k.conv3x3_nhwc_bf16(dy, w_rot, dx, 1)
+        else:
+            dx = torch.nn.grad.conv2d_input(
+                list(x.shape), w_nchw, dy, stride=stride, padding=1,
+            )
+        dw = torch.nn.grad.conv2d_weight(
+            x, list(w_nchw.shape), dy, stride=stride, padding=1,
+        ).to(ctx.c3_wdtype).permute(0, 1, 2, 3)
+        return dx, dw, None
+
+
+class MfmaConv3x3(nn.Conv2d):
+    """3x3 convolution on the hand-written MFMA implicit-GEMM kernel
+    (``conv_impl='mfma'``); falls back to the stock conv for inputs the
+    kernel doesn't cover (CPU, fp32, non-channels-last, Ci % 64 != 0)."""
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__(cin, cout, 3, stride=stride, padding=1, bias=False)
+
+    def forward(self, x):
+        if not (
+            x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and self.in_channels % 64 == 0
+        ):
+            return super().forward(x)
+        return _MfmaConv3x3Fn.apply(x, self.weight, self.stride[0])
 
 
 def _fused_supported(x: torch.Tensor, C: int) -> bool:

@@ -18,7 +18,12 @@ import torch
 import torch.nn as nn
 
 
-def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+def conv3x3(cin: int, cout: int, stride: int = 1,
+            impl: str = "miopen") -> nn.Conv2d:
+    if impl == "mfma" and cin % 64 == 0:
+        from .layers import MfmaConv3x3
+
+        return MfmaConv3x3(cin, cout, stride=stride)
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
@@ -43,8 +48,8 @@ class BasicBlock(nn.Module):
                  conv_impl="miopen"):
         super().__init__()
         self._fused = fused
-        self.conv1 = conv3x3(cin, planes, stride)
-        self.conv2 = conv3x3(planes, planes)
+        self.conv1 = conv3x3(cin, planes, stride, impl=conv_impl)
+        self.conv2 = conv3x3(planes, planes, impl=conv_impl)
         self.downsample = downsample
         self.stride = stride
         if fused:
@@ -81,7 +86,7 @@ class Bottleneck(nn.Module):
         super().__init__()
         self._fused = fused
         self.conv1 = conv1x1(cin, planes, impl=conv_impl)
-        self.conv2 = conv3x3(planes, planes, stride)
+        self.conv2 = conv3x3(planes, planes, stride, impl=conv_impl)
         self.conv3 = conv1x1(planes, planes * self.expansion,
                              impl=conv_impl)
         self.downsample = downsample

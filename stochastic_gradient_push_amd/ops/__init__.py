@@ -187,3 +187,53 @@ def sgd_step_(
             momentum_buf.mul_(momentum).add_(d, alpha=1.0 - dampening)
         d = d.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
     params.add_(d, alpha=-lr)
+
+
+# --------------------------------------------------------------- MFMA GEMM
+# Hand-written NT/TN bf16 GEMM dispatch (ops/csrc/gemm1x1_kernels.hip).
+# Variant choice per shape; see profiles/ for the measured ladder.
+
+#: use the 3-buffer barrier-crossing glds variant of the 256x128 kernel
+#: (set from hardware measurement; toggled by tools/gemm_bench.py results)
+V6_SPAN = True
+
+
+def gemm_nt(A: torch.Tensor, B: torch.Tensor,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ B[N,K]^T on the best hand-written MFMA kernel
+    for the shape (bf16 in/out, fp32 accumulate)."""
+    ext = _ext_for(A)
+    M, K = A.shape
+    N = B.shape[0]
+    C = out if out is not None else torch.empty(
+        M, N, device=A.device, dtype=torch.bfloat16
+    )
+    if K % 64 == 0:
+        if M % 256 == 0 and N % 128 == 0:
+            ext.gemm_nt_bf16_v6(A, B, C, span=V6_SPAN)
+        else:
+            ext.gemm_nt_bf16_v5(A, B, C)
+    else:
+        ext.gemm_nt_bf16_v2(A, B, C)
+    return C
+
+
+def gemm_tn_wgrad(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+    """dW[Co,Ci] = dy[M,Co]^T @ x[M,Ci] (fp32 out) via the split-M TN
+    kernel with deterministic partial reduction."""
+    ext = _ext_for(dy)
+    M, Co = dy.shape
+    Ci = x.shape[1]
+    co_tiles = (Co + 127) // 128
+    ci_tiles = (Ci + 127) // 128
+    # enough splits to fill the chip (>=2048 workgroups when M allows),
+    # bounded by the partials workspace (<= 128 MB)
+    split = max(1, 2048 // (co_tiles * ci_tiles))
+    split = min(split, max(1, M // 512))
+    max_split_mem = (128 << 20) // (Co * Ci * 4)
+    split = max(1, min(split, max_split_mem))
+    partials = torch.empty(split * Co * Ci, device=dy.device,
+                           dtype=torch.float32)
+    dw = torch.empty(Co * Ci, device=dy.device, dtype=torch.float32)
+    ext.gemm_tn_wgrad_bf16(dy, x, partials, dw, split)
+    return dw.view(Co, Ci)

@@ -63,6 +63,10 @@ void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
 void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
                             float* partials, float* dw, int64_t M, int Co,
                             int Ci, int split, hipStream_t s);
+void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
+                           ushort_t* Y, int Nb, int H, int W, int Ci,
+                           int Co, int Ho, int Wo, int stride,
+                           hipStream_t s);
 }
 
 namespace {
@@ -273,6 +277,36 @@ void gemm_nt_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
                    (int)K, current_stream(A));
 }
 
+// MFMA implicit-GEMM 3x3 conv (NHWC bf16, pad=1).  `x`/`y` are
+// channels_last 4-D activations (their memory IS NHWC); `w` is the
+// materialized [Co,3,3,Ci] weight.
+void conv3x3_nhwc_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor y,
+                       int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16
+              && x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "x must be bf16 channels_last");
+  TORCH_CHECK(y.is_cuda() && y.scalar_type() == torch::kBFloat16
+              && y.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "y must be bf16 channels_last");
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous()
+              && w.scalar_type() == torch::kBFloat16
+              && w.dim() == 4 && w.size(1) == 3 && w.size(2) == 3,
+              "w must be bf16 [Co,3,3,Ci] contiguous");
+  const int Nb = (int)x.size(0), Ci = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Co = (int)y.size(1), Ho = (int)y.size(2), Wo = (int)y.size(3);
+  TORCH_CHECK(w.size(0) == Co && w.size(3) == Ci, "weight shape mismatch");
+  TORCH_CHECK(Ci % 64 == 0, "Ci must be a multiple of 64");
+  TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
+  TORCH_CHECK(Ho == (H + 2 - 3) / stride + 1 && Wo == (W + 2 - 3) / stride + 1,
+              "output spatial mismatch (pad=1, 3x3)");
+  sgp_conv3x3_nhwc_bf16(
+      reinterpret_cast<const ushort_t*>(x.data_ptr()),
+      reinterpret_cast<const ushort_t*>(w.data_ptr()),
+      reinterpret_cast<ushort_t*>(y.data_ptr()), Nb, H, W, Ci, Co, Ho, Wo,
+      (int)stride, current_stream(x));
+}
+
 // ---------------------------------------------------------------- BN ops
 
 const ushort_t* bf16_ptr(const torch::Tensor& t) {
@@ -433,6 +467,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "barrier-crossing glds)");
   m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
         "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
+  m.def("conv3x3_nhwc_bf16", &conv3x3_nhwc_bf16, py::arg("x"),
+        py::arg("w"), py::arg("y"), py::arg("stride") = 1,
+        "MFMA implicit-GEMM 3x3 conv, NHWC bf16, pad=1");
   m.def("gemm_tn_wgrad_bf16", &gemm_tn_wgrad_bf16,
         "EXPERIMENTAL: dW = dy^T @ x with split-M partials (round-2 "
         "validation pending)");

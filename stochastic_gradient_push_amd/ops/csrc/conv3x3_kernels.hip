@@ -1,0 +1,228 @@
+// Hand-written MFMA implicit-GEMM 3x3 convolution — gfx950 (CDNA4).
+//
+// NHWC bf16, pad=1, stride 1 or 2, fp32 MFMA accumulation, bf16 out.
+// GEMM view (never materialized):  y[M, Co] = A[M, 9*Ci] @ W[Co, 9*Ci]^T
+// with M = N*Ho*Wo and A the im2col of x.  Requires Ci % 64 == 0 so a
+// BK=64 k-step stays inside one (r, s) filter tap — then every A-tile
+// row is a CONTIGUOUS 64-channel slice of x (vector loads, one
+// whole-row pad/bounds guard, no per-element guards — the guide's
+// trap 4c).  Weights are laid out [Co][3][3][Ci] (the channels_last
+// view of a PyTorch Conv2d weight), so B rows are contiguous too.
+//
+// Structure: 128x128 output tile, 256 threads = 4 waves in 2x2 (each
+// wave 64x64 = 4x4 fragments of v_mfma_f32_16x16x32_bf16), register-
+// staged double-buffered LDS with +8 bf16 row padding (the validated
+// v2 GEMM geometry, gemm1x1_kernels.hip), XCD-aware block->tile remap.
+// The m -> (n, ho, wo) decode happens ONCE per tile per thread (the
+// 4 staged rows are fixed), so the k-loop does only adds and guards.
+//
+// dgrad (stride 1) reuses THIS kernel: dx = conv3x3(dy, w_rot) with
+// w_rot[ci][r'][s'][co] = w[co][2-r'][2-s'][ci] (rotated + transposed,
+// materialized by the Python layer — it is only the small weight).
+//
+// Replaces MIOpen's igemm kernels on the ResNet-50 3x3 hot path
+// (reference workload: gossip_sgd.py:693-707 via torchvision/cuDNN).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace {
+
+typedef unsigned short ushort_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ ushort_t f2b_(float f) {
+  union { unsigned int i; float f; } v;
+  v.f = f;
+  unsigned int r = v.i + 0x7FFFu + ((v.i >> 16) & 1u);
+  return (ushort_t)(r >> 16);
+}
+
+#define CBM 128
+#define CBN 128
+#define CBK 64
+#define CSTRIDE 72  // 64 + 8 bf16 pad per LDS row
+
+__device__ __forceinline__ int64_t conv_xcd_bid() {
+  const int g8 = gridDim.x >> 3;
+  return (int64_t)(blockIdx.x & 7) * g8 + (blockIdx.x >> 3);
+}
+
+template <int STRIDE>
+__global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
+    const ushort_t* __restrict__ X,  // [N, H, W, Ci]
+    const ushort_t* __restrict__ Wt, // [Co, 3, 3, Ci]
+    ushort_t* __restrict__ Y,        // [N, Ho, Wo, Co]
+    int Nb, int H, int W, int Ci, int Co, int Ho, int Wo) {
+  __shared__ ushort_t As[2][CBM * CSTRIDE];
+  __shared__ ushort_t Bs[2][CBN * CSTRIDE];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;
+
+  const int64_t M = (int64_t)Nb * Ho * Wo;
+  const int K9 = 9 * Ci;
+  const int KT = K9 / CBK;
+
+  const int n_tiles = (Co + CBN - 1) / CBN;
+  const int64_t m_tiles = (M + CBM - 1) / CBM;
+  const int64_t total_tiles = m_tiles * n_tiles;
+
+  // staging assignment: 1024 segments of 8 bf16 per operand tile;
+  // thread t stages segs {t, t+256, t+512, t+768}: rows r8 = (t>>3) +
+  // {0,32,64,96}, column c8 = (t&7)*8
+  const int row0 = tid >> 3;
+  const int c8 = (tid & 7) * 8;
+
+  const int64_t bid0 = conv_xcd_bid();
+  for (int64_t tile = bid0; tile < total_tiles; tile += gridDim.x) {
+    const int64_t tm = (tile / n_tiles) * CBM;
+    const int tco = (int)(tile % n_tiles) * CBN;
+
+    // per-tile row decode (hoisted out of the k-loop)
+    int a_n[4], a_hi0[4], a_wi0[4];
+    bool a_ok[4];
+    const ushort_t* b_ptr[4];
+    bool b_ok[4];
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int row = row0 + s * 32;
+      const int64_t m = tm + row;
+      a_ok[s] = m < M;
+      if (a_ok[s]) {
+        const int64_t hw = (int64_t)Ho * Wo;
+        const int n = (int)(m / hw);
+        const int rem = (int)(m - (int64_t)n * hw);
+        const int ho = rem / Wo;
+        const int wo = rem - ho * Wo;
+        a_n[s] = n;
+        a_hi0[s] = ho * STRIDE - 1;
+        a_wi0[s] = wo * STRIDE - 1;
+      } else {
+        a_n[s] = 0; a_hi0[s] = -2; a_wi0[s] = -2;
+      }
+      const int co = tco + row;
+      b_ok[s] = co < Co;
+      b_ptr[s] = Wt + (int64_t)(b_ok[s] ? co : 0) * K9;
+    }
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    bf16x8 ra[4], rb[4];
+
+#define CONV_LOAD(kt)                                                      \
+  do {                                                                     \
+    const int kk = (kt)*CBK;                                               \
+    const int rs = kk / Ci;                                                \
+    const int ci0 = kk - rs * Ci;                                          \
+    const int r = rs / 3, sfs = rs - r * 3;                                \
+    _Pragma("unroll") for (int s = 0; s < 4; ++s) {                        \
+      const int hi = a_hi0[s] + r;                                         \
+      const int wi = a_wi0[s] + sfs;                                       \
+      const bool ok = a_ok[s] && (unsigned)hi < (unsigned)H                \
+                      && (unsigned)wi < (unsigned)W;                       \
+      if (ok) {                                                            \
+        ra[s] = *reinterpret_cast<const bf16x8*>(                          \
+            X + ((((int64_t)a_n[s] * H + hi) * W + wi) * Ci + ci0 + c8));  \
+      } else {                                                             \
+        ra[s] = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};                          \
+      }                                                                    \
+      if (b_ok[s]) {                                                       \
+        rb[s] = *reinterpret_cast<const bf16x8*>(b_ptr[s] + kk + c8);      \
+      } else {                                                             \
+        rb[s] = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};                          \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+#define CONV_WRITE(buf)                                                    \
+  do {                                                                     \
+    _Pragma("unroll") for (int s = 0; s < 4; ++s) {                        \
+      const int row = row0 + s * 32;                                       \
+      *reinterpret_cast<bf16x8*>(As[buf] + row * CSTRIDE + c8) = ra[s];    \
+      *reinterpret_cast<bf16x8*>(Bs[buf] + row * CSTRIDE + c8) = rb[s];    \
+    }                                                                      \
+  } while (0)
+
+    CONV_LOAD(0);
+    CONV_WRITE(0);
+
+    for (int kt = 0; kt < KT; ++kt) {
+      __syncthreads();
+      const int buf = kt & 1;
+      if (kt + 1 < KT) CONV_LOAD(kt + 1);  // issue early, hide behind MFMA
+
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          afrag[i] = *reinterpret_cast<const bf16x8*>(
+              As[buf] + (wm + i * 16 + frow) * CSTRIDE + ks * 32 + fk0);
+          bfrag[i] = *reinterpret_cast<const bf16x8*>(
+              Bs[buf] + (wn + i * 16 + frow) * CSTRIDE + ks * 32 + fk0);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      }
+
+      __syncthreads();
+      if (kt + 1 < KT) CONV_WRITE(buf ^ 1);
+    }
+#undef CONV_LOAD
+#undef CONV_WRITE
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const int64_t gm = tm + wm + i * 16 + (lane >> 4) * 4 + rr;
+          const int gc = tco + wn + j * 16 + (lane & 15);
+          if (gm < M && gc < Co) Y[gm * Co + gc] = f2b_(acc[i][j][rr]);
+        }
+  }
+}
+
+inline int conv_grid(int64_t M, int Co) {
+  int64_t tiles = ((M + CBM - 1) / CBM) * (int64_t)((Co + CBN - 1) / CBN);
+  if (tiles > 16384) tiles = 16384;
+  int g = (int)((tiles + 7) & ~7);  // multiple of 8 for the XCD remap
+  return g < 8 ? 8 : g;
+}
+
+}  // namespace
+
+extern "C" {
+
+void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
+                           ushort_t* Y, int Nb, int H, int W, int Ci,
+                           int Co, int Ho, int Wo, int stride,
+                           hipStream_t s) {
+  const int64_t M = (int64_t)Nb * Ho * Wo;
+  const int grid = conv_grid(M, Co);
+  if (stride == 1)
+    hipLaunchKernelGGL(k_conv3x3_nhwc_bf16<1>, dim3(grid), dim3(256), 0, s,
+                       X, Wt, Y, Nb, H, W, Ci, Co, Ho, Wo);
+  else
+    hipLaunchKernelGGL(k_conv3x3_nhwc_bf16<2>, dim3(grid), dim3(256), 0, s,
+                       X, Wt, Y, Nb, H, W, Ci, Co, Ho, Wo);
+}
+
+}  // extern "C"
