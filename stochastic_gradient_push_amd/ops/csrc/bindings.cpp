@@ -64,9 +64,9 @@ void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
                             float* partials, float* dw, int64_t M, int Co,
                             int Ci, int split, hipStream_t s);
 void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
-                           ushort_t* Y, int Nb, int H, int W, int Ci,
-                           int Co, int Ho, int Wo, int stride,
-                           hipStream_t s);
+                           ushort_t* Y, float* P, int Nb, int H, int W,
+                           int Ci, int Co, int Ho, int Wo, int stride,
+                           int split, hipStream_t s);
 }
 
 namespace {
@@ -250,7 +250,7 @@ void gemm_tn_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
               && partials.scalar_type() == torch::kFloat32
               && partials.numel() == split * Co * Ci,
               "partials must be fp32 [split*Co*Ci]");
-  TORCH_CHECK(dy.size(0) % 32 == 0, "M must be a multiple of 32");
+  TORCH_CHECK(dy.size(0) >= 1, "M must be positive");
   sgp_gemm_tn_wgrad_bf16(
       reinterpret_cast<const ushort_t*>(dy.data_ptr()),
       reinterpret_cast<const ushort_t*>(x.data_ptr()),
@@ -300,11 +300,33 @@ void conv3x3_nhwc_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor y,
   TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
   TORCH_CHECK(Ho == (H + 2 - 3) / stride + 1 && Wo == (W + 2 - 3) / stride + 1,
               "output spatial mismatch (pad=1, 3x3)");
+
+  // split-K to fill the chip on small-M shapes (e.g. the 7x7 stage has
+  // 52 tiles for ~512 resident-block slots).  Target ~768 blocks, keep
+  // >= 2 k-steps per split, and cap the fp32 partials workspace at
+  // 64 MB — the partial write+read traffic must stay small relative to
+  // the MACs or split-K loses (measured trade-off, profiles/r02_*)
+  const int64_t M = (int64_t)Nb * Ho * Wo;
+  const int64_t tiles = ((M + 127) / 128) * ((Co + 127) / 128);
+  const int KT = 9 * Ci / 64;
+  int64_t split = 768 / (tiles > 0 ? tiles : 1);
+  if (split > KT / 2) split = KT / 2;
+  const int64_t max_mem_split = (64ll << 20) / (M * Co * 4);
+  if (split > max_mem_split) split = max_mem_split;
+  if (split < 1) split = 1;
+
+  torch::Tensor partials;
+  float* pptr = nullptr;
+  if (split > 1) {
+    partials = torch::empty({split * M * Co},
+                            x.options().dtype(torch::kFloat32));
+    pptr = partials.data_ptr<float>();
+  }
   sgp_conv3x3_nhwc_bf16(
       reinterpret_cast<const ushort_t*>(x.data_ptr()),
       reinterpret_cast<const ushort_t*>(w.data_ptr()),
-      reinterpret_cast<ushort_t*>(y.data_ptr()), Nb, H, W, Ci, Co, Ho, Wo,
-      (int)stride, current_stream(x));
+      reinterpret_cast<ushort_t*>(y.data_ptr()), pptr, Nb, H, W, Ci, Co,
+      Ho, Wo, (int)stride, (int)split, current_stream(x));
 }
 
 // ---------------------------------------------------------------- BN ops

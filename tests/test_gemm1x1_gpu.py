@@ -221,7 +221,7 @@ def test_gemm_v5_throughput_readout():
 
 
 @pytest.mark.parametrize("span", [False, True])
-@pytest.mark.parametrize("shape", [(25088, 512, 512), (6272, 1024, 256),
+@pytest.mark.parametrize("shape", [(25088, 512, 512), (12800, 1024, 256),
                                    (100352, 256, 64), (512, 128, 64)])
 def test_gemm_nt_v6_matches_v1(shape, span):
     """v6 (256x128 tile, 8 waves; optional barrier-span glds) must be

@@ -161,6 +161,35 @@ def test_trainer_resume(tmp_path):
     _run_trainer(tmp_path, args + ["--resume", "True"])
 
 
+def test_trainer_mid_epoch_resume(tmp_path):
+    """Mid-epoch resume fast-forwards the sampler (gossip_sgd.py
+    train_epoch start_itr spoof; reference gossip_sgd.py:356-364):
+    after resuming from a checkpoint with itr=1, iteration 0 must NOT
+    be re-run and iteration 1 must be."""
+    import pandas as pd
+    import torch
+
+    args = ["--graph_type", "-1", "--master_port", "29817",
+            "--overwrite_checkpoints", "True", "--print_freq", "1"]
+    _run_trainer(tmp_path, args)
+
+    ckpt = f"{tmp_path}/ckpt/checkpoint_r0_n1.pth.tar"
+    state = torch.load(ckpt, weights_only=False)
+    assert state["epoch"] == 1 and state["itr"] == 0
+    state["epoch"] = 0
+    state["itr"] = 1
+    torch.save(state, ckpt)
+
+    _run_trainer(tmp_path, args + ["--resume", "True"])
+
+    csv = f"{tmp_path}/ckpt/out_r0_n1.csv"
+    df = pd.read_csv(csv)
+    train = df[df["itr"] >= 0]
+    # epoch 0, itr 1 ran twice (original + resumed); itr 0 only once
+    assert len(train[(train["Epoch"] == 0) & (train["itr"] == 1)]) == 2
+    assert len(train[(train["Epoch"] == 0) & (train["itr"] == 0)]) == 1
+
+
 def test_adpsgd_cli_two_rank(tmp_path):
     _run_trainer(
         tmp_path,
