@@ -71,7 +71,20 @@ class BilatGossipDataParallel(Module):
 
         if comm_device is None:
             comm_device = torch.device("cpu")
+        if comm_device.type == "cuda" and comm_device.index is None:
+            comm_device = torch.device("cuda", torch.cuda.current_device())
         self.__cpu_comm = comm_device.type == "cpu"
+        if not self.__cpu_comm:
+            # device-resident comm: shared CUDA tensors travel to the
+            # gossip process as dmabuf-IPC views of the SAME device
+            # memory (torch.multiprocessing CUDA sharing); p2p on device
+            # buffers needs RCCL — gloo only does host p2p
+            if (backend or "nccl") != "nccl":
+                raise ValueError(
+                    "comm_device=cuda requires backend='nccl' (RCCL): "
+                    "gloo cannot p2p device tensors"
+                )
+            backend = "nccl"
 
         self.dist_config = {
             "verbose": verbose,
@@ -96,7 +109,11 @@ class BilatGossipDataParallel(Module):
         # flat views over trainable params/grads in the trainer process
         self.flatp = FlatParams(module, flatten_grads=True)
 
-        # shared flat buffers handed to the gossip process
+        # shared flat buffers handed to the gossip process.  CPU: true
+        # shared memory.  CUDA: the mp.Queue transfer maps the SAME
+        # device allocation into the gossip process via dmabuf IPC
+        # (reference ad_psgd.py:72-74,105 only supported pinned CPU —
+        # this is the SURVEY C10 MI355X-native mode).
         shared_params = self.flatp.flat.detach().to(comm_device).clone()
         shared_grads = torch.zeros_like(shared_params)
         if self.__cpu_comm:
@@ -187,9 +204,15 @@ class BilatGossipDataParallel(Module):
 
     def _pull_model(self):
         """Copy the gossip process's current params into the module
-        (reference ad_psgd.py:220-229) — one flat copy."""
+        (reference ad_psgd.py:220-229) — one flat copy.
+
+        CUDA-comm: the mp.Lock orders HOST threads, not GPU streams in
+        two processes, so the read must complete before the lock is
+        released (the gossip process syncs its writes the same way)."""
         with self.gossip_lock:
             self.flatp.flat.copy_(self.gossip_params_flat)
+            if not self.__cpu_comm:
+                torch.cuda.current_stream().synchronize()
         return True
 
     def _transfer_grads(self):
@@ -197,6 +220,10 @@ class BilatGossipDataParallel(Module):
         ad_psgd.py:232-249)."""
         self.gossip_read_flag.wait()
         self.gossip_grads_flat.copy_(self.flatp.flat_grad)
+        if not self.__cpu_comm:
+            # the flag is host-side: the device copy must land before
+            # the gossip process is told the grads are ready
+            torch.cuda.current_stream().synchronize()
         self.gossip_read_flag.clear()
         self.train_write_flag.set()
         return True
@@ -218,6 +245,10 @@ class BilatGossipDataParallel(Module):
         RCCL/Gloo rank, applies grads with a fused SGD, and bilaterally
         averages with one peer per iteration."""
         with torch.no_grad():
+            if dist_config["comm_device"].type == "cuda":
+                # pin the gossip process to the trainer's GPU before the
+                # IPC tensors are materialized or RCCL initializes
+                torch.cuda.set_device(dist_config["comm_device"])
             gossip_params, gossip_grads = gossip_queue.get()
             momentum_buf = torch.zeros_like(gossip_params)
             first_step = True
@@ -272,6 +303,7 @@ class BilatGossipDataParallel(Module):
                     logger.debug(f"updated lr to {lr}")
                     gossip_update_flag.clear()
 
+                cuda_comm = dist_config["comm_device"].type == "cuda"
                 if train_write_flag.is_set():
                     bt = _time.time()
                     with gossip_lock:
@@ -284,6 +316,8 @@ class BilatGossipDataParallel(Module):
                             first_step=first_step,
                         )
                         first_step = False
+                        if cuda_comm:
+                            torch.cuda.current_stream().synchronize()
                     train_write_flag.clear()
                     gossip_read_flag.set()
                     model_meter.update(_time.time() - bt)
@@ -293,12 +327,16 @@ class BilatGossipDataParallel(Module):
                     bt = _time.time()
                     with gossip_lock:
                         out_msg = gossip_params.clone()
+                        if cuda_comm:
+                            torch.cuda.current_stream().synchronize()
                     in_msg, completed = gossiper.mix(out_msg)
                     if not isinstance(completed, bool) or completed:
                         with gossip_lock:
                             ops.average_(
                                 gossip_params, in_msg.to(gossip_params.device)
                             )
+                            if cuda_comm:
+                                torch.cuda.current_stream().synchronize()
                     gossip_meter.update(_time.time() - bt)
                     logger.debug(gossip_meter)
                 except RuntimeError as e:

@@ -140,3 +140,16 @@ def test_adpsgd_update_lr():
         p.join(timeout=180)
     for p in procs:
         assert p.exitcode == 0
+
+
+def test_cuda_comm_requires_nccl():
+    """comm_device=cuda must reject gloo (device p2p needs RCCL)."""
+    from stochastic_gradient_push_amd import BilatGossipDataParallel
+
+    with pytest.raises(ValueError):
+        BilatGossipDataParallel(
+            tiny_model(0),
+            master_addr="127.0.0.1", master_port=12345,
+            backend="gloo", world_size=2, rank=0,
+            comm_device=torch.device("cuda", 0),
+        )
