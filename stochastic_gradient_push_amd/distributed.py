@@ -48,6 +48,7 @@ from .gossiper import PushPull, PushSum
 from .graphs import NPeerDynamicDirectedExponentialGraph as NPDDEGraph
 from .mixing import UniformMixing
 from .ops.flat import FlatParams
+from .utils.roctx import trace as _roctx
 from .utils.helpers import (
     communicate,
     create_process_group,
@@ -484,20 +485,21 @@ class GossipDataParallel(Module):
                 self.transfer_params(mix=False)
                 return False
 
-            self.lazy_ps_factor.copy_(self.gossip_ps_factor)
-            self.ps_numerator()
-            self.ps_weight += self.gossip_ps_weight
-            if self.lazy_mixing:
-                self.ps_weight *= self.lazy_ps_factor
-                ops.add_scale_cast_(
-                    self.flatp.flat,
-                    self.gossip_device_buffer,
-                    self.lazy_ps_factor.to(self.flatp.flat.dtype),
-                )
-            else:
-                ops.add_scale_cast_(
-                    self.flatp.flat, self.gossip_device_buffer, 1.0
-                )
+            with _roctx("sgp:merge_received"):
+                self.lazy_ps_factor.copy_(self.gossip_ps_factor)
+                self.ps_numerator()
+                self.ps_weight += self.gossip_ps_weight
+                if self.lazy_mixing:
+                    self.ps_weight *= self.lazy_ps_factor
+                    ops.add_scale_cast_(
+                        self.flatp.flat,
+                        self.gossip_device_buffer,
+                        self.lazy_ps_factor.to(self.flatp.flat.dtype),
+                    )
+                else:
+                    ops.add_scale_cast_(
+                        self.flatp.flat, self.gossip_device_buffer, 1.0
+                    )
 
             self.logger.debug(f"updated ps-weight {self.ps_weight}")
             self.gossip_flag.clear()
@@ -521,6 +523,8 @@ class GossipDataParallel(Module):
 
         mix = mix and not self.lazy_mixing
 
+        _rc = _roctx("sgp:transfer_params")
+        _rc.__enter__()
         self.ps_numerator()
         if mix:
             self.ps_weight *= self.gossip_ps_factor
@@ -556,6 +560,7 @@ class GossipDataParallel(Module):
         self.params_mixed = False
         self.gossiping = True
         self.train_flag.set()
+        _rc.__exit__(None, None, None)
         return True
 
     # -- gossip thread -------------------------------------------------------

@@ -32,6 +32,7 @@ import torch.distributed as dist
 
 from .graphs import GraphManager
 from .mixing import MixingManager, UniformMixing
+from .utils.roctx import trace as _roctx
 
 
 class _PendingRecv:
@@ -352,8 +353,9 @@ class PushSum(Gossiper):
             self.logger.debug(
                 f"in/out -peers {self.in_edges}/{self.out_edges}"
             )
-        send = self._prep_out_msg(out_msg, ps_weight, residual)
-        self._exchange(send, residual)
+        with _roctx("sgp:gossip_mix"):
+            send = self._prep_out_msg(out_msg, ps_weight, residual)
+            self._exchange(send, residual)
         self.refresh_peers_()
         # re-derive weights for the NEW peer set so mixing_weights['lo']
         # (read by the wrapper as next round's pre-scale factor) matches

@@ -18,6 +18,7 @@ import torch
 
 from . import sgd_step_
 from .flat import FlatParams
+from ..utils.roctx import trace as _roctx
 
 
 class FusedSGD:
@@ -91,17 +92,18 @@ class FusedSGD:
         ):
             self.flatp.rewire_grads()
         self.sync_lr()
-        sgd_step_(
-            self.flatp.flat,
-            self.flatp.flat_grad,
-            self.momentum_buf,
-            lr=self._lr_dev if self._lr_dev is not None else g["lr"],
-            momentum=g["momentum"],
-            weight_decay=g["weight_decay"],
-            dampening=g["dampening"],
-            nesterov=g["nesterov"],
-            first_step=self._first_step,
-        )
+        with _roctx("sgp:fused_sgd_step"):
+            sgd_step_(
+                self.flatp.flat,
+                self.flatp.flat_grad,
+                self.momentum_buf,
+                lr=self._lr_dev if self._lr_dev is not None else g["lr"],
+                momentum=g["momentum"],
+                weight_decay=g["weight_decay"],
+                dampening=g["dampening"],
+                nesterov=g["nesterov"],
+                first_step=self._first_step,
+            )
         self._first_step = False
 
     def state_dict(self):

@@ -120,3 +120,50 @@ def test_plotting_renders_png(tmp_path):
     assert os.path.exists(out)
     out2 = plot_scaling({1: 0.5, 2: 0.3, 4: 0.2}, f"{tmp_path}/scale.png")
     assert os.path.exists(out2)
+
+
+def test_experiment_matrix_and_plot(tmp_path):
+    """ETH/IB experiment-tag matrix (reference plotting.py:55-134
+    equivalent): matrix shape is consistent and plot_matrix renders,
+    skipping absent experiments and including present ones."""
+    import shutil
+    import subprocess
+    import sys
+
+    from visualization.plotting import experiment_matrix, plot_matrix
+
+    for kind, n_groups in (("eth", 3), ("ib", 2), ("transformer", 2)):
+        nodes, fpaths, tags, legends, colors = experiment_matrix(kind)
+        assert len(fpaths) == len(tags) == len(legends) == len(colors) \
+            == n_groups
+        for t, l, c in zip(tags, legends, colors):
+            assert len(t) == len(l) == len(c) == len(nodes)
+
+    # produce one real run and place it under an ETH tag
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update(RANK="0", WORLD_SIZE="1", MASTER_ADDR="127.0.0.1")
+    subprocess.run(
+        [
+            sys.executable, os.path.join(repo, "gossip_sgd.py"),
+            "--num_epochs", "1", "--num_iterations_per_training_epoch", "4",
+            "--batch_size", "2", "--synthetic_size", "16",
+            "--model", "resnet18", "--num_classes", "10",
+            "--image_size", "32", "--num_dataloader_workers", "0",
+            "--device", "cpu", "--checkpoint_dir", f"{tmp_path}/ck/",
+            "--num_itr_ignore", "0", "--graph_type", "-1",
+            "--train_fast", "True", "--print_freq", "1",
+            "--master_port", "29822",
+        ],
+        env=env, check=True, timeout=240, cwd=str(tmp_path),
+    )
+    res = tmp_path / "results_dir" / "out_files"
+    res.mkdir(parents=True)
+    shutil.copy(
+        f"{tmp_path}/ck/out_r0_n1.csv", res / "SGP-4ETHout_r0_n4.csv"
+    )
+    out = plot_matrix(
+        "eth", f"{tmp_path}/eth.png",
+        results_dir=str(tmp_path / "results_dir"), metric="avg:Loss",
+    )
+    assert os.path.exists(out)

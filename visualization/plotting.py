@@ -135,6 +135,92 @@ def plot_scaling(results, out_path):
     return out_path
 
 
+def experiment_matrix(kind, results_dir="results_dir"):
+    """Experiment-tag matrix for the transport comparison plots
+    (reference plotting.py:55-134 get_eth_config/get_ib_config/
+    get_transformer_config): per algorithm, the CSV path template,
+    the per-node-count experiment tags, legend strings and a colormap
+    ramp.  Tags follow the launch scripts' checkpoint-dir naming
+    (``{tag}out_r{r}_n{n}.csv``).
+    """
+    import numpy as np
+    from matplotlib import cm
+
+    def ramp(cmap, n):
+        return [cmap(x) for x in np.linspace(0.3, 0.8, n)]
+
+    nodes = [4, 8, 16, 32]
+    fpath = results_dir + "/out_files/{tag}out_r{r}_n{n}.csv"
+    if kind == "ib":
+        groups = [
+            ("SGP", ["SGP-%dIB" % n for n in nodes], cm.Blues),
+            ("AR-SGD", ["AR-%dIB" % n for n in nodes], cm.Reds),
+        ]
+    elif kind == "eth":
+        groups = [
+            ("AR-SGD", ["AR-%dETH" % n for n in nodes], cm.Reds),
+            ("D-PSGD", ["DPSGD-%dETH" % n for n in nodes], cm.Greens),
+            ("SGP", ["SGP-%dETH" % n for n in nodes], cm.Blues),
+        ]
+    elif kind == "transformer":
+        nodes = [8, 8]
+        fpath = results_dir + "/transformer_{tag}_test.out"
+        groups = [
+            ("SGP", ["ps_sm", "ps"], cm.Blues),
+            ("SGD", ["ar_sm", "ar"], cm.Reds),
+        ]
+    else:
+        raise ValueError(f"unknown experiment matrix '{kind}'")
+
+    fpaths, tags, legends, colors = [], [], [], []
+    for name, group_tags, cmap in groups:
+        fpaths.append(fpath)
+        tags.append(group_tags)
+        if kind == "transformer":
+            legends.append([f"{name} (25K batch)", f"{name} (400K batch)"])
+        else:
+            legends.append([f"{name} {n} nodes" for n in nodes])
+        colors.append(ramp(cmap, len(group_tags)))
+    return nodes, fpaths, tags, legends, colors
+
+
+def plot_matrix(kind, out_path, results_dir="results_dir",
+                metric="avg:Prec@1"):
+    """Overlay every algorithm's runs from the tag matrix, one curve per
+    (algorithm, node-count): metric vs wall-clock training time
+    (reference plotting.py's ETH/IB comparison figures).  Experiments
+    whose CSVs are absent are skipped (the matrix describes the full
+    grid; partial result dirs are normal)."""
+    plt = _plt()
+    nodes, fpaths, tags, legends, colors = experiment_matrix(
+        kind, results_dir
+    )
+    fig, ax = plt.subplots()
+    plotted = 0
+    for fpath, group_tags, group_legends, group_colors in zip(
+        fpaths, tags, legends, colors
+    ):
+        for tag, legend, color, n in zip(
+            group_tags, group_legends, group_colors, nodes
+        ):
+            path = fpath.format(tag=tag, r=0, n=n)
+            if not os.path.exists(path):
+                continue
+            try:
+                meta, train_df, _ = parse_csv(path)
+            except Exception:
+                continue
+            x = train_df["BT(s)"].cumsum()
+            ax.plot(x, train_df[metric], label=legend, color=color)
+            plotted += 1
+    ax.set_xlabel("training time (s)")
+    ax.set_ylabel(metric)
+    if plotted:
+        ax.legend()
+    fig.savefig(out_path, bbox_inches="tight")
+    return out_path
+
+
 def plot_transformer(df, out_path):
     """Validation NLL vs optimizer steps (reference plotting.py:231)."""
     plt = _plt()
