@@ -211,7 +211,18 @@ def gemm_nt(A: torch.Tensor, B: torch.Tensor,
     # K=64 (KT=1) degenerates the glds pipelines (measured: v2 40 TF vs
     # v3 26 TF on 100352x64x64) -> register-staged v2 below K=128
     if K % 64 == 0 and K >= 128:
-        if M % 256 == 0 and N % 128 == 0:
+        tiles = ((M + 127) // 128) * ((N + 127) // 128)
+        split = 0
+        if tiles < 384 and K >= 512:
+            # under-filled launch (layer4 1x1s, M = 1568): split the K
+            # reduction to recover chip fill, bounded by the fp32
+            # partials workspace (64 MB) so the extra partial traffic
+            # stays small next to the MACs
+            split = min(64, max(2, 768 // tiles), K // 128,
+                        (64 << 20) // (M * N * 4))
+        if split >= 2:
+            ext.gemm_nt_splitk_bf16(A, B, C, split)
+        elif M % 256 == 0 and N % 128 == 0:
             ext.gemm_nt_bf16_v6(A, B, C, span=V6_SPAN)
         else:
             ext.gemm_nt_bf16_v5(A, B, C)
@@ -228,10 +239,11 @@ def gemm_tn_wgrad(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     Ci = x.shape[1]
     co_tiles = (Co + 127) // 128
     ci_tiles = (Ci + 127) // 128
-    # enough splits to fill the chip (>=2048 workgroups when M allows),
-    # bounded by the partials workspace (<= 128 MB)
-    split = max(1, 2048 // (co_tiles * ci_tiles))
-    split = min(split, max(1, M // 512))
+    # enough splits to fill the chip (~768 blocks), bounded by the
+    # reduce kernel's depth (deep strided partial sums turn the reduce
+    # into the bottleneck — measured at split 2048) and the workspace
+    split = max(1, 768 // (co_tiles * ci_tiles))
+    split = min(split, 64, max(1, M // 512))
     max_split_mem = (128 << 20) // (Co * Ci * 4)
     split = max(1, min(split, max_split_mem))
     partials = torch.empty(split * Co * Ci, device=dy.device,

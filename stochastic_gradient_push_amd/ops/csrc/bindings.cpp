@@ -60,6 +60,9 @@ void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, int span, hipStream_t s);
+void sgp_gemm_nt_splitk_bf16(const ushort_t* A, const ushort_t* B,
+                             float* P, ushort_t* C, int64_t M, int N,
+                             int K, int split, hipStream_t s);
 void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
                             float* partials, float* dw, int64_t M, int Co,
                             int Ci, int split, hipStream_t s);
@@ -223,6 +226,22 @@ void gemm_nt_bf16_v6(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                       current_stream(A));
 }
 
+void gemm_nt_splitk_bf16(torch::Tensor A, torch::Tensor B,
+                         torch::Tensor C, int64_t split) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(1) % 64 == 0, "split-K requires K %% 64 == 0");
+  TORCH_CHECK(split >= 1 && split <= 64);
+  const int64_t M = A.size(0), N = B.size(0);
+  torch::Tensor partials = torch::empty(
+      {split * M * N}, A.options().dtype(torch::kFloat32));
+  sgp_gemm_nt_splitk_bf16(
+      reinterpret_cast<const ushort_t*>(A.data_ptr()),
+      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+      partials.data_ptr<float>(),
+      reinterpret_cast<ushort_t*>(C.data_ptr()), M, (int)N,
+      (int)A.size(1), (int)split, current_stream(A));
+}
+
 void gemm_nt_bf16_v4(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
   gemm_nt_check(A, B, C);
   TORCH_CHECK(A.size(0) % 128 == 0 && B.size(0) % 128 == 0
@@ -309,7 +328,12 @@ void conv3x3_nhwc_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor y,
   const int64_t M = (int64_t)Nb * Ho * Wo;
   const int64_t tiles = ((M + 127) / 128) * ((Co + 127) / 128);
   const int KT = 9 * Ci / 64;
-  int64_t split = 768 / (tiles > 0 ? tiles : 1);
+  // only under-filled launches with small outputs benefit: the fp32
+  // partial write+read traffic scales with split*M*Co (measured: 28x28
+  // shape 239 -> 176 TF with split, M*Co = 3.2M; 7x7 shape 54 -> 187
+  // TF, M*Co = 0.8M)
+  const bool want_split = tiles < 384 && M * Co <= (2ll << 20);
+  int64_t split = want_split ? 768 / (tiles > 0 ? tiles : 1) : 1;
   if (split > KT / 2) split = KT / 2;
   const int64_t max_mem_split = (64ll << 20) / (M * Co * 4);
   if (split > max_mem_split) split = max_mem_split;
@@ -487,6 +511,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("C"), py::arg("span") = false,
         "NT bf16 MFMA GEMM v6 (256x128 tile, 8 waves; span=3-buf "
         "barrier-crossing glds)");
+  m.def("gemm_nt_splitk_bf16", &gemm_nt_splitk_bf16, py::arg("A"),
+        py::arg("B"), py::arg("C"), py::arg("split"),
+        "NT bf16 MFMA GEMM, K split over block groups (small-M shapes)");
   m.def("gemm_nt_bf16_v4", &gemm_nt_bf16_v4,
         "3-buffer glds, raw barrier + counted vmcnt (full tiles only)");
   m.def("conv3x3_nhwc_bf16", &conv3x3_nhwc_bf16, py::arg("x"),

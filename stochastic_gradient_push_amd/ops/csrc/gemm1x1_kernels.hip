@@ -615,6 +615,156 @@ __global__ __launch_bounds__(256) void k_gemm_nt_bf16_v4(
 #undef GLDS4
 }
 
+// ----------------------------------------------- gemm NT split-K (v2s)
+// v2's register-staged 128x128 geometry with the K reduction split over
+// `split` block groups writing fp32 partials [split][M][N] (reduced to
+// bf16 by k_gemm_nt_reduce).  For under-filled shapes (the 7x7-stage
+// 1x1 convs: M = 1568 -> 13 m-tiles) where neither glds pipeline can
+// fill the chip.
+__global__ __launch_bounds__(256) void k_gemm_nt_splitk_bf16(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    float* __restrict__ P, int64_t M, int N, int K, int split) {
+  constexpr int BKT = 64;
+  constexpr int STRIDE = BKT + 8;
+  constexpr int SEGS = BM * BKT / 8 / 256;
+  __shared__ ushort_t As[2][BM * STRIDE];
+  __shared__ ushort_t Bs[2][BN * STRIDE];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;
+
+  constexpr int SEG_PER_ROW = BKT / 8;
+  int rr[SEGS], cc[SEGS];
+#pragma unroll
+  for (int s = 0; s < SEGS; ++s) {
+    const int seg = tid + s * 256;
+    rr[s] = seg / SEG_PER_ROW;
+    cc[s] = (seg % SEG_PER_ROW) * 8;
+  }
+
+  const int n_tiles = (N + BN - 1) / BN;
+  const int64_t m_tiles = (M + BM - 1) / BM;
+  const int64_t total = m_tiles * n_tiles * split;
+  const int KT_all = K / BKT;
+
+  const int64_t bid0 = xcd_virtual_bid();
+  for (int64_t t = bid0; t < total; t += gridDim.x) {
+    const int ks = (int)(t % split);
+    const int64_t ct = t / split;
+    const int64_t tm = (ct / n_tiles) * BM;
+    const int tn = (int)(ct % n_tiles) * BN;
+    const int kt_lo = (int)(((int64_t)KT_all * ks) / split);
+    const int kt_hi = (int)(((int64_t)KT_all * (ks + 1)) / split);
+    const int KT = kt_hi - kt_lo;
+    if (KT <= 0) continue;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    bf16x8 ra[SEGS], rb[SEGS];
+    const bool full = (tm + BM <= M) && (tn + BN <= N);
+
+#define LOAD_T(k0)                                                         \
+  do {                                                                     \
+    if (full) {                                                            \
+      _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                   \
+        ra[s] = *reinterpret_cast<const bf16x8*>(                          \
+            A + (tm + rr[s]) * K + (k0) + cc[s]);                          \
+        rb[s] = *reinterpret_cast<const bf16x8*>(                          \
+            B + (int64_t)(tn + rr[s]) * K + (k0) + cc[s]);                 \
+      }                                                                    \
+    } else {                                                               \
+      _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                   \
+        const int64_t ga = tm + rr[s];                                     \
+        const int gb = tn + rr[s];                                         \
+        ushort_t ta[8], tb[8];                                             \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j) {                    \
+          ta[j] = (ga < M) ? A[ga * K + (k0) + cc[s] + j] : (ushort_t)0;   \
+          tb[j] = (gb < N) ? B[(int64_t)gb * K + (k0) + cc[s] + j]         \
+                           : (ushort_t)0;                                  \
+        }                                                                  \
+        ra[s] = *reinterpret_cast<bf16x8*>(ta);                            \
+        rb[s] = *reinterpret_cast<bf16x8*>(tb);                            \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+#define WRITE_T(buf)                                                       \
+  do {                                                                     \
+    _Pragma("unroll") for (int s = 0; s < SEGS; ++s) {                     \
+      *reinterpret_cast<bf16x8*>(As[buf] + rr[s] * STRIDE + cc[s]) =       \
+          ra[s];                                                           \
+      *reinterpret_cast<bf16x8*>(Bs[buf] + rr[s] * STRIDE + cc[s]) =       \
+          rb[s];                                                           \
+    }                                                                      \
+  } while (0)
+
+    LOAD_T((int64_t)kt_lo * BKT);
+    WRITE_T(0);
+
+    for (int kt = 0; kt < KT; ++kt) {
+      __syncthreads();
+      const int buf = kt & 1;
+      if (kt + 1 < KT) LOAD_T((int64_t)(kt_lo + kt + 1) * BKT);
+
+#pragma unroll
+      for (int kk = 0; kk < BKT / 32; ++kk) {
+        bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          afrag[i] = *reinterpret_cast<const bf16x8*>(
+              As[buf] + (wm + i * 16 + frow) * STRIDE + kk * 32 + fk0);
+          bfrag[i] = *reinterpret_cast<const bf16x8*>(
+              Bs[buf] + (wn + i * 16 + frow) * STRIDE + kk * 32 + fk0);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      }
+
+      __syncthreads();
+      if (kt + 1 < KT) WRITE_T(buf ^ 1);
+    }
+#undef LOAD_T
+#undef WRITE_T
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+          const int gc = tn + wn + j * 16 + (lane & 15);
+          if (gr < M && gc < N)
+            P[((int64_t)ks * M + gr) * N + gc] = acc[i][j][r];
+        }
+  }
+}
+
+__global__ void k_gemm_nt_reduce_bf16(const float* __restrict__ P,
+                                      ushort_t* __restrict__ C,
+                                      int64_t numel, int split) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    float acc = 0.f;
+    for (int k = 0; k < split; ++k) acc += P[(int64_t)k * numel + i];
+    C[i] = f2b(acc);
+  }
+}
+
 // ------------------------------------- gemm v6 (256x128 tile, 8 waves)
 // The 128x128 tile's LDS-staging traffic (2 x 16 KB per 64-K-step for
 // 4.2 MFLOP -> 66 flops/byte) caps it below hipBLASLt.  256x128 with
@@ -757,14 +907,21 @@ __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
   }
 }
 
-// ------------------------------------------- wgrad TN (EXPERIMENTAL)
+// --------------------------------------------------------- wgrad TN
 // dW[Co,Ci] = sum_m dy[m,co] * x[m,ci] — the 1x1-conv weight gradient.
 // K = M is huge, so blocks split the M range and store fp32 partials
 // [split][128][128]; a second kernel reduces them (deterministic, no
 // atomics — same pattern as the BN reductions).  Staging transposes on
 // write into the v1 kernel's padded [c][m] LDS image, so the MFMA inner
-// loop is identical to the NT kernel's.  Compile-verified this round;
-// hardware validation is round-2 work (tests marked gpu_experimental).
+// loop is identical to the NT kernel's.
+//
+// Staging geometry (round-2 rewrite after the first hardware numbers
+// came in 3-6x below roofline): lanes are M-FASTEST — 16-lane groups
+// cover 16 consecutive m-rows of one 8-channel column group, so the
+// global read is ONE bf16x8 vector load per lane (rows Co*2 B apart,
+// no per-element guards) and the 8 transposed LDS writes land on 8
+// consecutive dwords per 16-lane group (2-way conflict) instead of the
+// old c-fastest mapping's single bank (16-way).
 __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
     const ushort_t* __restrict__ dy,  // [M, Co]
     const ushort_t* __restrict__ x,   // [M, Ci]
@@ -785,7 +942,8 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
   const int ci_tiles = (Ci + BN - 1) / BN;
   const int64_t tiles = (int64_t)co_tiles * ci_tiles * split;
 
-  for (int64_t t = blockIdx.x; t < tiles; t += gridDim.x) {
+  const int64_t bid0 = xcd_virtual_bid();
+  for (int64_t t = bid0; t < tiles; t += gridDim.x) {
     const int s = (int)(t % split);
     const int64_t ct = t / split;
     const int tco = (int)(ct / ci_tiles) * BM;
@@ -800,43 +958,58 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
 #pragma unroll
       for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+    // seg -> (m-row, channel-group), m-fastest within 16-lane groups
+    const int mrow = tid & 31;         // 32 m-rows per BK step
+    const int cgrp = tid >> 5;         // 8 channel-groups per 256 thr
+    const bool a_full = tco + BM <= Co;
+    const bool b_full = tci + BN <= Ci;
+
     for (int64_t k0 = m0; k0 < m1; k0 += BK) {
       __syncthreads();
-      // stage BK=32 m-rows x 128 channels per operand, transposing into
-      // the padded [channel][m] image; 2 segments x 2 operands/thread
+      const int64_t gm = k0 + mrow;
+      const bool mok = gm < m1;
+      // 16 channel-groups of 8 per operand row: this thread covers
+      // cgrp and cgrp+8
 #pragma unroll
-      for (int seg4 = 0; seg4 < 2; ++seg4) {
-        const int seg = tid + seg4 * 256;  // 0..511
-        const int mrow = seg >> 4;         // 32 m-rows
-        const int c8 = (seg & 15) * 8;     // 16 channel-groups of 8
-        const int64_t gm = k0 + mrow;
+      for (int half = 0; half < 2; ++half) {
+        const int c8 = (cgrp + half * 8) * 8;
         // dy -> As
         {
-          ushort_t tmp[8];
-          const bool ok = gm < m1;
+          bf16x8 v;
+          const int gc0 = tco + c8;
+          if (mok && a_full) {
+            v = *reinterpret_cast<const bf16x8*>(dy + gm * Co + gc0);
+          } else {
+            ushort_t tmp[8];
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int gc = tco + c8 + j;
-            tmp[j] = (ok && gc < Co)
-                ? dy[gm * Co + gc] : (ushort_t)0;
+            for (int j = 0; j < 8; ++j)
+              tmp[j] = (mok && gc0 + j < Co)
+                  ? dy[gm * Co + gc0 + j] : (ushort_t)0;
+            v = *reinterpret_cast<bf16x8*>(tmp);
           }
+          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            As[(c8 + j) * LDS_STRIDE + mrow] = tmp[j];
+            As[(c8 + j) * LDS_STRIDE + mrow] = e[j];
         }
         // x -> Bs
         {
-          ushort_t tmp[8];
-          const bool ok = gm < m1;
+          bf16x8 v;
+          const int gc0 = tci + c8;
+          if (mok && b_full) {
+            v = *reinterpret_cast<const bf16x8*>(x + gm * Ci + gc0);
+          } else {
+            ushort_t tmp[8];
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int gc = tci + c8 + j;
-            tmp[j] = (ok && gc < Ci)
-                ? x[gm * Ci + gc] : (ushort_t)0;
+            for (int j = 0; j < 8; ++j)
+              tmp[j] = (mok && gc0 + j < Ci)
+                  ? x[gm * Ci + gc0 + j] : (ushort_t)0;
+            v = *reinterpret_cast<bf16x8*>(tmp);
           }
+          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            Bs[(c8 + j) * LDS_STRIDE + mrow] = tmp[j];
+            Bs[(c8 + j) * LDS_STRIDE + mrow] = e[j];
         }
       }
       __syncthreads();
@@ -920,6 +1093,23 @@ void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
                      A, B, C, M, N, K);
 }
 
+void sgp_gemm_nt_splitk_bf16(const ushort_t* A, const ushort_t* B,
+                             float* P, ushort_t* C, int64_t M, int N,
+                             int K, int split, hipStream_t s) {
+  const int64_t tiles =
+      ((M + BM - 1) / BM) * (int64_t)((N + BN - 1) / BN) * split;
+  int grid = (int)(tiles > 16384 ? 16384 : tiles);
+  grid = (grid + 7) & ~7;
+  if (grid < 8) grid = 8;
+  hipLaunchKernelGGL(k_gemm_nt_splitk_bf16, dim3(grid), dim3(256), 0, s,
+                     A, B, P, M, N, K, split);
+  const int64_t numel = M * N;
+  int rgrid = (int)(((numel + 255) / 256) > 8192 ? 8192
+                                                 : (numel + 255) / 256);
+  hipLaunchKernelGGL(k_gemm_nt_reduce_bf16, dim3(rgrid < 1 ? 1 : rgrid),
+                     dim3(256), 0, s, P, C, numel, split);
+}
+
 void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, int span, hipStream_t s) {
   int64_t tiles = (M / BM6) * (int64_t)(N / BN6);
@@ -946,6 +1136,7 @@ void sgp_gemm_tn_wgrad_bf16(const ushort_t* dy, const ushort_t* x,
   const int64_t tiles =
       (int64_t)((Co + 127) / 128) * ((Ci + 127) / 128) * split;
   int grid = (int)(tiles > 16384 ? 16384 : tiles);
+  grid = (grid + 7) & ~7;  // multiple of 8: XCD remap bijectivity
   hipLaunchKernelGGL(k_gemm_tn_partial_bf16, dim3(grid), dim3(256), 0, s,
                      dy, x, partials, M, Co, Ci, split);
   const int64_t numel = (int64_t)Co * Ci;

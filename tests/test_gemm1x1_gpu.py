@@ -264,6 +264,50 @@ def test_gemm_v6_throughput_readout():
           f"{flops / t6 / 1e12:.1f} vs {flops / t6s / 1e12:.1f} TF")
 
 
+@pytest.mark.parametrize("shape,split", [
+    ((1568, 512, 1024), 8), ((1568, 2048, 512), 4),
+    ((640, 128, 512), 2), ((130, 72, 512), 3),
+])
+def test_gemm_nt_splitk_matches_matmul(shape, split):
+    M, N, K = shape
+    torch.manual_seed(7)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_splitk_bf16(A, B, C, split)
+    torch.cuda.synchronize()
+    ref = A.float() @ B.float().t()
+    err = (C.float() - ref).abs()
+    scale = ref.abs().mean() + 1e-3
+    assert (err.mean() / scale) < 5e-2, (err.mean() / scale).item()
+    assert torch.allclose(C.float(), ref, atol=3.0, rtol=8e-2), (
+        err.max().item()
+    )
+
+
+def test_gemm_splitk_throughput_readout():
+    M, N, K = 1568, 512, 1024
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+
+    def t(f, n=30):
+        for _ in range(5):
+            f()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n
+
+    flops = 2.0 * M * N * K
+    t2 = t(lambda: ext().gemm_nt_bf16_v2(A, B, C))
+    tk = t(lambda: ext().gemm_nt_splitk_bf16(A, B, C, 8))
+    print(f"\n[gemm 1568x512x1024 v2 vs splitk8] {flops / t2 / 1e12:.1f} "
+          f"vs {flops / tk / 1e12:.1f} TF")
+
+
 def test_gemm_nt_v4_matches_v1_and_throughput():
     M, N, K = 25088, 512, 512
     torch.manual_seed(4)

@@ -57,7 +57,7 @@ def main():
         (25088, 512, 512),         # the ladder's reference shape
     ]
 
-    cols = ("v2", "v3", "v5", "v6", "v6s", "lib")
+    cols = ("v2", "v3", "v5", "v6", "v6s", "auto", "lib")
     print(f"{'M':>8} {'N':>5} {'K':>5} | " + " ".join(
         f"{c:>6}" for c in cols) + "  (TFLOP/s)")
     for M, N, K in shapes:
@@ -77,23 +77,20 @@ def main():
                     A, B, C, span=False)) / 1e12
                 r["v6s"] = fl / timeit(lambda: ext.gemm_nt_bf16_v6(
                     A, B, C, span=True)) / 1e12
+        r["auto"] = fl / timeit(lambda: ops.gemm_nt(A, B, out=C)) / 1e12
         r["lib"] = fl / timeit(lambda: torch.matmul(A, B.t())) / 1e12
         print(f"{M:>8} {N:>5} {K:>5} | " + " ".join(
             f"{r[k]:>6.0f}" for k in cols))
 
-    # wgrad TN: dW[Co,Ci] = dy^T @ x on the same shapes (Co=N, Ci=K)
-    print(f"\nwgrad TN (split={args.wgrad_split}):")
+    # wgrad TN: dW[Co,Ci] = dy^T @ x on the same shapes (Co=N, Ci=K);
+    # "tn" uses the dispatcher's split heuristic
+    print("\nwgrad TN (dispatcher split):")
     print(f"{'M':>8} {'Co':>5} {'Ci':>5} | {'tn':>6} {'lib':>6}  (TFLOP/s)")
     for M, Co, Ci in shapes:
         dy = torch.randn(M, Co, device=dev).to(torch.bfloat16)
         x = torch.randn(M, Ci, device=dev).to(torch.bfloat16)
-        split = args.wgrad_split
-        partials = torch.zeros(split * Co * Ci, device=dev)
-        dw = torch.zeros(Co * Ci, device=dev)
         fl = 2.0 * M * Co * Ci
-        t_tn = fl / timeit(
-            lambda: ext.gemm_tn_wgrad_bf16(dy, x, partials, dw, split)
-        ) / 1e12
+        t_tn = fl / timeit(lambda: ops.gemm_tn_wgrad(dy, x)) / 1e12
         t_lib = fl / timeit(lambda: torch.matmul(dy.t(), x)) / 1e12
         print(f"{M:>8} {Co:>5} {Ci:>5} | {t_tn:>6.0f} {t_lib:>6.0f}")
 
