@@ -63,6 +63,12 @@ def parse_args():
         choices=["bf16", "fp32"],
         help="wire format of gossip messages (bf16 halves xGMI bytes)",
     )
+    p.add_argument(
+        "--master-weights", type=str, default="auto",
+        choices=["auto", "on", "off"],
+        help="bf16 working weights with fp32 master (removes per-layer "
+             "autocast weight casts; auto = on for bf16+fused on GPU)",
+    )
     return p.parse_args()
 
 
@@ -99,6 +105,7 @@ def main():
     if cuda and not args.no_channels_last:
         model = model.to(memory_format=torch.channels_last)
 
+    use_master = False
     use_ddp = args.algorithm == "ar" and world_size > 1
     if use_ddp:
         model = torch.nn.parallel.DistributedDataParallel(
@@ -121,6 +128,10 @@ def main():
         # captured compute runs), with the lazy sync state machine; the
         # wrapper-level overlap hooks are only for the eager fallback.
         graph_planned = cuda and not args.no_graph
+        use_master = args.master_weights == "on" or (
+            args.master_weights == "auto"
+            and cuda and args.dtype == "bf16" and args.opt == "fused"
+        )
         gdp = GossipDataParallel(
             model,
             graph=graph,
@@ -132,6 +143,7 @@ def main():
                 torch.bfloat16 if args.gossip_dtype == "bf16" else None
             ),
             flatten_grads=(args.opt == "fused"),
+            working_dtype=torch.bfloat16 if use_master else None,
         )
         model = gdp
         if args.opt == "fused":
@@ -278,6 +290,7 @@ def main():
                 "norm": args.norm,
                 "gossip_dtype": args.gossip_dtype,
                 "conv_impl": args.conv_impl,
+                "master_weights": use_master,
             },
         }
         print(json.dumps(result), flush=True)

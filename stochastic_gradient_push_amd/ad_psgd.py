@@ -367,10 +367,7 @@ class BilatGossipDataParallel(Module):
     def __make_backward_hook(self):
         def hook(*unused):
             # keep grads wired to the flat buffer
-            p0 = self.flatp.params[0]
-            if p0.grad is None or (
-                p0.grad.data_ptr() != self.flatp.flat_grad.data_ptr()
-            ):
+            if not self.flatp.grads_wired():
                 self.flatp.rewire_grads()
             if self.gossip_enable:
                 self._transfer_grads()

@@ -81,6 +81,7 @@ class GossipDataParallel(Module):
         flatten_grads: bool = True,
         gossip_dtype: Optional[torch.dtype] = None,
         comm_backend: str = "c10d",
+        working_dtype: Optional[torch.dtype] = None,
     ):
         super().__init__()
 
@@ -174,8 +175,13 @@ class GossipDataParallel(Module):
         self.asynch = synch_freq > 0
         self.logger = make_logger(rank, verbose)
 
-        # flat parameter/grad views — the central data structure
-        self.flatp = FlatParams(module, flatten_grads=flatten_grads)
+        # flat parameter/grad views — the central data structure.
+        # working_dtype=bfloat16 keeps an fp32 MASTER here (gossip and
+        # the optimizer see only the master) while the model computes
+        # with bf16 shadow weights (see FlatParams docstring).
+        self.flatp = FlatParams(
+            module, flatten_grads=flatten_grads, working_dtype=working_dtype
+        )
         self._flatten_grads = flatten_grads
 
         # push-sum state
@@ -318,14 +324,31 @@ class GossipDataParallel(Module):
         distributed.py:209-222)."""
         if finish_gossip:
             self._query_gossip_queue()
-        return {
+        out = {
             "state_dict": super().state_dict(),
             "ps_weight": self.ps_weight.cpu(),
             "is_ps_numerator": self.is_ps_numerator,
         }
+        if self.flatp.shadow is not None:
+            # module weights above are the bf16 working copies; keep the
+            # fp32 master so resume is bit-exact
+            out["master_flat"] = self.flatp.flat.detach().cpu()
+        return out
 
     def load_state_dict(self, load_dict):
         super().load_state_dict(load_dict["state_dict"])
+        if self.flatp.shadow is not None:
+            if "master_flat" in load_dict:
+                self.flatp.flat.copy_(
+                    load_dict["master_flat"].to(self.flatp.flat.device)
+                )
+            else:
+                # bf16-only checkpoint (e.g. from a non-master run):
+                # upcast the loaded working weights into the master
+                self.flatp.flat.narrow(0, 0, self.flatp.n_cast).copy_(
+                    self.flatp.shadow
+                )
+            self.flatp.sync_shadow()
         self.ps_weight = load_dict["ps_weight"].to(
             device=self.dist_config["comm_device"]
         )
@@ -427,6 +450,7 @@ class GossipDataParallel(Module):
         distributed.py:278-296); the flat param broadcast is one op."""
         src = self.dist_config["rank"] * self.nprocs_per_node
         dist.broadcast(self.flatp.flat, src=src, group=self.local_node_group)
+        self.flatp.sync_shadow()
         buffers = [b.data for b in self.module.buffers()]
         if buffers:
             communicate(
@@ -446,6 +470,7 @@ class GossipDataParallel(Module):
                     self.flatp.flat,
                     self.ps_weight.to(self.flatp.flat.dtype),
                 )
+                self.flatp.sync_shadow()
             self.is_ps_numerator = True
 
     def unbias(self):
@@ -456,6 +481,7 @@ class GossipDataParallel(Module):
                     self.flatp.flat,
                     (1.0 / self.ps_weight).to(self.flatp.flat.dtype),
                 )
+                self.flatp.sync_shadow()
             self.is_ps_numerator = False
 
     def _query_gossip_queue(self, non_blocking: bool = False):
@@ -501,6 +527,7 @@ class GossipDataParallel(Module):
                         self.flatp.flat, self.gossip_device_buffer, 1.0
                     )
 
+            self.flatp.sync_shadow()
             self.logger.debug(f"updated ps-weight {self.ps_weight}")
             self.gossip_flag.clear()
             self.params_mixed = True
@@ -543,6 +570,8 @@ class GossipDataParallel(Module):
             ops.pack_mix_cast_(
                 self.flatp.flat, self.gossip_device_buffer, 1.0
             )
+        if mix:
+            self.flatp.sync_shadow()
 
         if self._cuda:
             # hand the staging buffer to the gossip stream; async copy to
@@ -700,10 +729,12 @@ class GossipDataParallel(Module):
                 # all-reduce when grads are flat (reference
                 # distributed.py:551-562 flattened per dtype each step)
                 if self._flatten_grads and self.flatp.flat_grad is not None:
-                    self.flatp.flat_grad.div_(self.nprocs_per_node)
-                    dist.all_reduce(
-                        self.flatp.flat_grad, group=self.local_node_group
-                    )
+                    bufs = [self.flatp.flat_grad]
+                    if self.flatp.flat_grad_w is not None:
+                        bufs.append(self.flatp.flat_grad_w)
+                    for buf in bufs:
+                        buf.div_(self.nprocs_per_node)
+                        dist.all_reduce(buf, group=self.local_node_group)
                 else:
                     grads = [
                         p.grad.data
@@ -732,10 +763,7 @@ class GossipDataParallel(Module):
             if self._flatten_grads:
                 # re-wire grads if an optimizer's zero_grad(set_to_none)
                 # detached them from the flat buffer
-                p0 = self.flatp.params[0]
-                if p0.grad is None or (
-                    p0.grad.data_ptr() != self.flatp.flat_grad.data_ptr()
-                ):
+                if not self.flatp.grads_wired():
                     self.flatp.rewire_grads()
             if self.gossip_enable:
                 non_blocking = self.num_updates < self.synch_freq

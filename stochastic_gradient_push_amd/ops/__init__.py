@@ -208,26 +208,36 @@ def gemm_nt(A: torch.Tensor, B: torch.Tensor,
     C = out if out is not None else torch.empty(
         M, N, device=A.device, dtype=torch.bfloat16
     )
-    # K=64 (KT=1) degenerates the glds pipelines (measured: v2 40 TF vs
-    # v3 26 TF on 100352x64x64) -> register-staged v2 below K=128
-    if K % 64 == 0 and K >= 128:
-        tiles = ((M + 127) // 128) * ((N + 127) // 128)
+    # Variant choice fitted to the measured ladder over every ResNet-50
+    # 1x1 shape (profiles/r02_gemm_bench3.txt):
+    #   * K not 64-aligned               -> v2 (only kernel without the
+    #     K%64 constraint)
+    #   * small M (layer4, M=1568): chip under-fill dominates -> split-K
+    #     when K is deep enough to amortize the fp32 partial traffic,
+    #     else register-staged v2 (glds pipelines lose at 13 m-tiles)
+    #   * N=64 (layer1): half the 128-wide tile is waste; v2 degrades
+    #     least
+    #   * N>=256 with 256|M: the 256x128 v6 tile (more flops per staged
+    #     byte) wins; otherwise the 128x128 glds v5
+    if K % 64 != 0:
+        ext.gemm_nt_bf16_v2(A, B, C)
+        return C
+    tiles = ((M + 127) // 128) * ((N + 127) // 128)
+    if M <= 2048:
         split = 0
-        if tiles < 384 and K >= 512:
-            # under-filled launch (layer4 1x1s, M = 1568): split the K
-            # reduction to recover chip fill, bounded by the fp32
-            # partials workspace (64 MB) so the extra partial traffic
-            # stays small next to the MACs
+        if K >= 1024 and tiles <= 256:
             split = min(64, max(2, 768 // tiles), K // 128,
                         (64 << 20) // (M * N * 4))
         if split >= 2:
             ext.gemm_nt_splitk_bf16(A, B, C, split)
-        elif M % 256 == 0 and N % 128 == 0:
-            ext.gemm_nt_bf16_v6(A, B, C, span=V6_SPAN)
         else:
-            ext.gemm_nt_bf16_v5(A, B, C)
-    else:
+            ext.gemm_nt_bf16_v2(A, B, C)
+    elif N < 128:
         ext.gemm_nt_bf16_v2(A, B, C)
+    elif N >= 256 and M % 256 == 0 and N % 128 == 0:
+        ext.gemm_nt_bf16_v6(A, B, C, span=V6_SPAN)
+    else:
+        ext.gemm_nt_bf16_v5(A, B, C)
     return C
 
 
@@ -243,7 +253,7 @@ def gemm_tn_wgrad(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     # reduce kernel's depth (deep strided partial sums turn the reduce
     # into the bottleneck — measured at split 2048) and the workspace
     split = max(1, 768 // (co_tiles * ci_tiles))
-    split = min(split, 64, max(1, M // 512))
+    split = min(split, 64, max(1, M // 128))
     max_split_mem = (128 << 20) // (Co * Ci * 4)
     split = max(1, min(split, max_split_mem))
     partials = torch.empty(split * Co * Ci, device=dy.device,
@@ -251,3 +261,42 @@ def gemm_tn_wgrad(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     dw = torch.empty(Co * Ci, device=dy.device, dtype=torch.float32)
     ext.gemm_tn_wgrad_bf16(dy, x, partials, dw, split)
     return dw.view(Co, Ci)
+
+
+def cast_shadow_(master: torch.Tensor, shadow: torch.Tensor) -> None:
+    """shadow = bf16(master) in one pass (working-weight refresh)."""
+    if master.is_cuda:
+        _ext_for(master).cast_shadow_(master, shadow)
+    else:
+        shadow.copy_(master)
+
+
+def sgd_step_bf16gs_(
+    params: torch.Tensor,
+    grads_bf16: torch.Tensor,
+    momentum_buf: torch.Tensor,
+    shadow: torch.Tensor,
+    lr,
+    momentum: float = 0.0,
+    weight_decay: float = 0.0,
+    dampening: float = 0.0,
+    nesterov: bool = False,
+    first_step: bool = False,
+) -> None:
+    """Fused SGD over the fp32 master with bf16 gradients; refreshes the
+    bf16 working-weight shadow in the same pass."""
+    if params.is_cuda:
+        _ext_for(params).sgd_step_bf16gs_(
+            params, grads_bf16, momentum_buf, shadow,
+            _as_scalar_tensor(lr, params), float(momentum),
+            float(weight_decay), float(dampening), bool(nesterov),
+            bool(first_step),
+        )
+        return
+    # CPU oracle: upcast grads, run the reference path, recast shadow
+    sgd_step_(
+        params, grads_bf16.float(), momentum_buf, lr,
+        momentum=momentum, weight_decay=weight_decay, dampening=dampening,
+        nesterov=nesterov, first_step=first_step,
+    )
+    shadow.copy_(params)

@@ -43,6 +43,12 @@ void sgp_pack_mix_bf16(float* x, unsigned short* out, const float* a,
                        int64_t n, hipStream_t stream);
 void sgp_add_scale_bf16(float* x, const unsigned short* r, const float* a,
                         int64_t n, hipStream_t stream);
+void sgp_sgd_step_bf16gs(float* p, const unsigned short* g, float* buf,
+                         unsigned short* shadow, const float* lr_ptr,
+                         double mu, double wd, double damp, bool nesterov,
+                         bool first, int64_t n, hipStream_t stream);
+void sgp_cast_shadow(const float* p, unsigned short* shadow, int64_t n,
+                     hipStream_t stream);
 void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
                   double mu, double wd, double damp, bool nesterov,
                   bool first, int64_t n, hipStream_t stream);
@@ -136,6 +142,41 @@ void sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
   sgp_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
                lr.data_ptr<float>(), momentum, weight_decay, dampening,
                nesterov, first_step, p.numel(), current_stream(p));
+}
+
+void sgd_step_bf16gs_(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
+                      torch::Tensor shadow, torch::Tensor lr,
+                      double momentum, double weight_decay,
+                      double dampening, bool nesterov, bool first_step) {
+  check_flat(p, "p");
+  check_flat(buf, "buf");
+  check_scalar(lr, p);
+  TORCH_CHECK(g.is_cuda() && g.is_contiguous()
+              && g.scalar_type() == torch::kBFloat16,
+              "g must be contiguous bf16");
+  TORCH_CHECK(shadow.is_cuda() && shadow.is_contiguous()
+              && shadow.scalar_type() == torch::kBFloat16,
+              "shadow must be contiguous bf16");
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == buf.numel()
+              && p.numel() == shadow.numel(), "size mismatch");
+  sgp_sgd_step_bf16gs(
+      p.data_ptr<float>(),
+      reinterpret_cast<const unsigned short*>(g.data_ptr()),
+      buf.data_ptr<float>(),
+      reinterpret_cast<unsigned short*>(shadow.data_ptr()),
+      lr.data_ptr<float>(), momentum, weight_decay, dampening, nesterov,
+      first_step, p.numel(), current_stream(p));
+}
+
+void cast_shadow_(torch::Tensor p, torch::Tensor shadow) {
+  check_flat(p, "p");
+  TORCH_CHECK(shadow.is_cuda() && shadow.is_contiguous()
+              && shadow.scalar_type() == torch::kBFloat16,
+              "shadow must be contiguous bf16");
+  TORCH_CHECK(p.numel() == shadow.numel(), "size mismatch");
+  sgp_cast_shadow(p.data_ptr<float>(),
+                  reinterpret_cast<unsigned short*>(shadow.data_ptr()),
+                  p.numel(), current_stream(p));
 }
 
 void pack_mix_bf16_(torch::Tensor x, torch::Tensor out, torch::Tensor a) {
@@ -498,6 +539,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_mix_", &pack_mix_, "x *= a; out = x");
   m.def("average_", &average_, "x = (x + y) / 2");
   m.def("sgd_step_", &sgd_step_, "fused momentum-SGD step");
+  m.def("sgd_step_bf16gs_", &sgd_step_bf16gs_,
+        "fused momentum-SGD step, bf16 grads + bf16 shadow write-back");
+  m.def("cast_shadow_", &cast_shadow_, "fp32 master -> bf16 shadow");
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 MFMA layout probe");
   m.def("gemm_nt_bf16", &gemm_nt_bf16,
         "C[M,N] = A[M,K] @ B[N,K]^T, bf16 MFMA, fp32 accumulate");

@@ -224,6 +224,82 @@ __global__ void k_sgd(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+// SGD with bf16 gradients and a bf16 working-weight shadow: the fp32
+// master is updated from bf16 grads (mixed-precision recipe) and the
+// bf16 copy the model computes with is refreshed in the same pass —
+// removes every per-layer autocast weight-cast and grad-cast kernel
+// from the step.
+template <bool kMomentum, bool kNesterov, bool kFirst>
+__global__ void k_sgd_bf16gs(float* __restrict__ p,
+                             const unsigned short* __restrict__ g,
+                             float* __restrict__ buf,
+                             unsigned short* __restrict__ shadow,
+                             const float* __restrict__ lr_ptr, float mu,
+                             float wd, float damp, int64_t n4, int64_t n) {
+  const float lr = *lr_ptr;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n4; i += stride) {
+    float4 pv = ld4(p + 4 * i);
+    const ushort4 gu = *reinterpret_cast<const ushort4*>(g + 4 * i);
+    float d0 = b2f_(gu.x) + wd * pv.x, d1 = b2f_(gu.y) + wd * pv.y,
+          d2 = b2f_(gu.z) + wd * pv.z, d3 = b2f_(gu.w) + wd * pv.w;
+    if (kMomentum) {
+      float4 bv;
+      if (kFirst) {
+        bv = make_float4(d0, d1, d2, d3);
+      } else {
+        bv = ld4(buf + 4 * i);
+        bv.x = mu * bv.x + (1.f - damp) * d0;
+        bv.y = mu * bv.y + (1.f - damp) * d1;
+        bv.z = mu * bv.z + (1.f - damp) * d2;
+        bv.w = mu * bv.w + (1.f - damp) * d3;
+      }
+      st4(buf + 4 * i, bv);
+      if (kNesterov) {
+        d0 += mu * bv.x; d1 += mu * bv.y; d2 += mu * bv.z; d3 += mu * bv.w;
+      } else {
+        d0 = bv.x; d1 = bv.y; d2 = bv.z; d3 = bv.w;
+      }
+    }
+    pv.x -= lr * d0; pv.y -= lr * d1; pv.z -= lr * d2; pv.w -= lr * d3;
+    st4(p + 4 * i, pv);
+    ushort4 sv;
+    sv.x = f2b_(pv.x); sv.y = f2b_(pv.y);
+    sv.z = f2b_(pv.z); sv.w = f2b_(pv.w);
+    *reinterpret_cast<ushort4*>(shadow + 4 * i) = sv;
+  }
+  for (int64_t j = 4 * n4 + (blockIdx.x * blockDim.x + threadIdx.x);
+       j < n; j += stride) {
+    float d = b2f_(g[j]) + wd * p[j];
+    if (kMomentum) {
+      float b = kFirst ? d : mu * buf[j] + (1.f - damp) * d;
+      buf[j] = b;
+      d = kNesterov ? d + mu * b : b;
+    }
+    p[j] -= lr * d;
+    shadow[j] = f2b_(p[j]);
+  }
+}
+
+// fp32 -> bf16 shadow refresh (after gossip merges touch the master)
+__global__ void k_cast_shadow(const float* __restrict__ p,
+                              unsigned short* __restrict__ shadow,
+                              int64_t n4, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n4; i += stride) {
+    const float4 pv = ld4(p + 4 * i);
+    ushort4 sv;
+    sv.x = f2b_(pv.x); sv.y = f2b_(pv.y);
+    sv.z = f2b_(pv.z); sv.w = f2b_(pv.w);
+    *reinterpret_cast<ushort4*>(shadow + 4 * i) = sv;
+  }
+  for (int64_t j = 4 * n4 + (blockIdx.x * blockDim.x + threadIdx.x);
+       j < n; j += stride)
+    shadow[j] = f2b_(p[j]);
+}
+
 inline int grid_for(int64_t work) {
   int64_t blocks = (work + THREADS - 1) / THREADS;
   if (blocks < 1) blocks = 1;
@@ -300,6 +376,38 @@ void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
     LAUNCH(false, false, false);
   }
 #undef LAUNCH
+}
+
+void sgp_sgd_step_bf16gs(float* p, const unsigned short* g, float* buf,
+                         unsigned short* shadow, const float* lr_ptr,
+                         double mu, double wd, double damp, bool nesterov,
+                         bool first, int64_t n, hipStream_t stream) {
+  const int64_t n4 = n / 4;
+  const dim3 grid(grid_for(n4 ? n4 : n));
+  const float muf = (float)mu, wdf = (float)wd, dampf = (float)damp;
+#define LAUNCH(M, N, F)                                                   \
+  hipLaunchKernelGGL((k_sgd_bf16gs<M, N, F>), grid, dim3(THREADS), 0,     \
+                     stream, p, g, buf, shadow, lr_ptr, muf, wdf, dampf,  \
+                     n4, n)
+  if (mu != 0.0) {
+    if (nesterov) {
+      if (first) LAUNCH(true, true, true);
+      else LAUNCH(true, true, false);
+    } else {
+      if (first) LAUNCH(true, false, true);
+      else LAUNCH(true, false, false);
+    }
+  } else {
+    LAUNCH(false, false, false);
+  }
+#undef LAUNCH
+}
+
+void sgp_cast_shadow(const float* p, unsigned short* shadow, int64_t n,
+                     hipStream_t stream) {
+  const int64_t n4 = n / 4;
+  hipLaunchKernelGGL(k_cast_shadow, dim3(grid_for(n4 ? n4 : n)),
+                     dim3(THREADS), 0, stream, p, shadow, n4, n);
 }
 
 }  // extern "C"

@@ -64,9 +64,23 @@ class FlatParams:
     buffer and points every ``p.grad`` at a view of it, so autograd
     accumulates directly into the contiguous buffer and the optimizer can
     run one fused kernel over the whole model.
+
+    ``working_dtype=torch.bfloat16`` enables the mixed-precision master-
+    weight layout: the flat buffer stays the fp32 MASTER (the optimizer
+    and the push-sum gossip operate on it unchanged), but the matrix-
+    shaped parameters (``ndim >= 2`` — convs and linears; BN scale/shift
+    stay fp32 for the fused BN kernels) are re-pointed at a bf16 SHADOW
+    of the master's leading section.  The model then computes directly
+    with bf16 weights — removing every per-layer autocast weight-cast
+    kernel — and their gradients arrive in a bf16 flat gradient buffer
+    that the fused SGD consumes directly (ops.sgd_step_ bf16-grad
+    variant refreshes the shadow in the same pass).  Parameter ORDER in
+    the flat buffer is repartitioned (cast set first), identically on
+    every rank, so gossip stays coherent.
     """
 
-    def __init__(self, module: torch.nn.Module, flatten_grads: bool = False):
+    def __init__(self, module: torch.nn.Module, flatten_grads: bool = False,
+                 working_dtype: Optional[torch.dtype] = None):
         params = [p for p in module.parameters() if p.requires_grad]
         assert len(params) > 0, "module has no trainable parameters"
         dtypes = {p.dtype for p in params}
@@ -74,24 +88,85 @@ class FlatParams:
             f"FlatParams supports a single param dtype, got {dtypes}; "
             "use one FlatParams per dtype"
         )
+        self.working_dtype = working_dtype
+        if working_dtype is not None:
+            assert working_dtype == torch.bfloat16
+            assert flatten_grads, "working_dtype requires flatten_grads"
+            cast = [p for p in params if p.ndim >= 2]
+            keep = [p for p in params if p.ndim < 2]
+            params = cast + keep
+            self.n_cast = sum(p.numel() for p in cast)
+        else:
+            self.n_cast = 0
         self.params = params
-        self._buf = FlatBuffer([p.data for p in params])
-        # re-point parameter storages at the flat views
-        for p, v in zip(params, self._buf.views):
-            p.data = v
+        self._buf = FlatBuffer([p.data.float() if working_dtype is not None
+                                else p.data for p in params])
         self.flat: torch.Tensor = self._buf.flat
 
+        self.shadow: Optional[torch.Tensor] = None
+        if working_dtype is not None:
+            self.shadow = torch.empty(
+                self.n_cast, dtype=working_dtype, device=self.flat.device
+            )
+            self.shadow.copy_(self.flat.narrow(0, 0, self.n_cast))
+        # re-point parameter storages: cast set at shadow views, the
+        # rest at fp32 master views
+        offset = 0
+        for p, v in zip(params, self._buf.views):
+            if self.shadow is not None and offset < self.n_cast:
+                p.data = self.shadow.narrow(
+                    0, offset, p.numel()
+                ).view_as(p)
+            else:
+                p.data = v
+            offset += p.numel()
+
         self.flat_grad: Optional[torch.Tensor] = None
+        self.flat_grad_w: Optional[torch.Tensor] = None
         if flatten_grads:
-            self.flat_grad = torch.zeros_like(self.flat)
-            offset = 0
-            for p in params:
-                n = p.numel()
-                p.grad = self.flat_grad.narrow(0, offset, n).view_as(p)
-                offset += n
+            if self.shadow is not None:
+                # bf16 grads for the cast set, fp32 for the rest
+                self.flat_grad_w = torch.zeros(
+                    self.n_cast, dtype=working_dtype,
+                    device=self.flat.device,
+                )
+                self.flat_grad = torch.zeros(
+                    self.flat.numel() - self.n_cast,
+                    dtype=self.flat.dtype, device=self.flat.device,
+                )
+            else:
+                self.flat_grad = torch.zeros_like(self.flat)
+            self.rewire_grads()
 
     def numel(self) -> int:
         return self.flat.numel()
+
+    def _grad_view(self, offset: int, n: int):
+        if self.shadow is not None and offset < self.n_cast:
+            return self.flat_grad_w.narrow(0, offset, n)
+        base = offset - self.n_cast
+        return self.flat_grad.narrow(0, base, n)
+
+    def grads_wired(self) -> bool:
+        """True when the first parameter's ``.grad`` still aliases the
+        flat gradient buffer (the cheap per-step wiring check)."""
+        if self.flat_grad is None:
+            return True
+        p0 = self.params[0]
+        ref = (self.flat_grad_w
+               if self.flat_grad_w is not None and self.n_cast
+               else self.flat_grad)
+        return p0.grad is not None and p0.grad.data_ptr() == ref.data_ptr()
+
+    def sync_shadow(self) -> None:
+        """Refresh the bf16 working weights from the fp32 master (call
+        after anything other than the fused SGD mutates the master —
+        gossip merges, load_state_dict, bias/de-bias rescales)."""
+        if self.shadow is None:
+            return
+        from . import cast_shadow_
+
+        cast_shadow_(self.flat.narrow(0, 0, self.n_cast), self.shadow)
 
     def rewire_grads(self) -> None:
         """Re-point ``p.grad`` at the flat grad views.
@@ -107,7 +182,7 @@ class FlatParams:
         offset = 0
         for p in self.params:
             n = p.numel()
-            g = self.flat_grad.narrow(0, offset, n).view_as(p)
+            g = self._grad_view(offset, n).view_as(p)
             if p.grad is None:
                 g.detach().zero_()
                 p.grad = g
@@ -119,4 +194,6 @@ class FlatParams:
     def zero_grad(self) -> None:
         assert self.flat_grad is not None
         self.flat_grad.zero_()
+        if self.flat_grad_w is not None:
+            self.flat_grad_w.zero_()
         self.rewire_grads()

@@ -39,9 +39,13 @@ class FusedSGD:
             "FusedSGD requires FlatParams(flatten_grads=True)"
         )
         self.flatp = flatp
-        self.momentum_buf = (
-            torch.zeros_like(flatp.flat) if momentum != 0.0 else flatp.flat_grad
-        )
+        self._mixed = flatp.shadow is not None
+        if momentum != 0.0:
+            self.momentum_buf = torch.zeros_like(flatp.flat)
+        elif self._mixed:
+            self.momentum_buf = torch.zeros_like(flatp.flat)
+        else:
+            self.momentum_buf = flatp.flat_grad
         self._first_step = True
         # device-resident lr: the step kernel reads it by pointer, so a
         # hipGraph-captured step sees live schedule updates (sync_lr())
@@ -86,24 +90,54 @@ class FusedSGD:
     def step(self, closure=None):
         g = self.param_groups[0]
         # autograd may have detached grads (e.g. someone else's zero_grad)
-        p0 = self.flatp.params[0]
-        if p0.grad is None or (
-            p0.grad.data_ptr() != self.flatp.flat_grad.data_ptr()
-        ):
+        if not self.flatp.grads_wired():
             self.flatp.rewire_grads()
         self.sync_lr()
+        lr = self._lr_dev if self._lr_dev is not None else g["lr"]
         with _roctx("sgp:fused_sgd_step"):
-            sgd_step_(
-                self.flatp.flat,
-                self.flatp.flat_grad,
-                self.momentum_buf,
-                lr=self._lr_dev if self._lr_dev is not None else g["lr"],
-                momentum=g["momentum"],
-                weight_decay=g["weight_decay"],
-                dampening=g["dampening"],
-                nesterov=g["nesterov"],
-                first_step=self._first_step,
-            )
+            if self._mixed:
+                # cast-set section: fp32 master <- bf16 grads, bf16
+                # working-weight shadow refreshed in the same kernel
+                from . import sgd_step_bf16gs_
+
+                n_cast = self.flatp.n_cast
+                sgd_step_bf16gs_(
+                    self.flatp.flat.narrow(0, 0, n_cast),
+                    self.flatp.flat_grad_w,
+                    self.momentum_buf.narrow(0, 0, n_cast),
+                    self.flatp.shadow,
+                    lr=lr,
+                    momentum=g["momentum"],
+                    weight_decay=g["weight_decay"],
+                    dampening=g["dampening"],
+                    nesterov=g["nesterov"],
+                    first_step=self._first_step,
+                )
+                rest = self.flatp.flat.numel() - n_cast
+                if rest:
+                    sgd_step_(
+                        self.flatp.flat.narrow(0, n_cast, rest),
+                        self.flatp.flat_grad,
+                        self.momentum_buf.narrow(0, n_cast, rest),
+                        lr=lr,
+                        momentum=g["momentum"],
+                        weight_decay=g["weight_decay"],
+                        dampening=g["dampening"],
+                        nesterov=g["nesterov"],
+                        first_step=self._first_step,
+                    )
+            else:
+                sgd_step_(
+                    self.flatp.flat,
+                    self.flatp.flat_grad,
+                    self.momentum_buf,
+                    lr=lr,
+                    momentum=g["momentum"],
+                    weight_decay=g["weight_decay"],
+                    dampening=g["dampening"],
+                    nesterov=g["nesterov"],
+                    first_step=self._first_step,
+                )
         self._first_step = False
 
     def state_dict(self):

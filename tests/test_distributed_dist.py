@@ -238,3 +238,62 @@ def _bf16_wire_consensus(rank, world_size):
 
 def test_bf16_wire_consensus():
     run_dist(_bf16_wire_consensus, world_size=2)
+
+
+def _master_weight_consensus(rank, world_size):
+    """working_dtype=bf16 (fp32 master + bf16 working weights): gossip
+    operates on the master; consensus at zero lr, and the working
+    weights track the master after every merge."""
+    from stochastic_gradient_push_amd import GossipDataParallel
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    model = tiny_model(seed=rank)
+    cast = [p for p in model.parameters() if p.ndim >= 2]
+    keep = [p for p in model.parameters() if p.ndim < 2]
+    flat0 = torch.cat([p.detach().reshape(-1) for p in cast + keep])
+    target = flat0.clone()
+    dist.all_reduce(target)
+    target /= world_size
+
+    gdp = GossipDataParallel(
+        model, push_sum=True, working_dtype=torch.bfloat16,
+    )
+    opt = FusedSGD(gdp.flatp, lr=0.0)
+    loss_fn = nn.CrossEntropyLoss()
+    gdp.train()
+    x = torch.randn(2, 3, 8, 8)
+    y = torch.randint(0, 10, (2,))
+    for _ in range(30):
+        with torch.autocast(device_type="cpu", dtype=torch.bfloat16):
+            loss = loss_fn(gdp(x), y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        gdp.transfer_params()
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    gdp.unbias()
+    flat = gdp.flatp.flat.detach()
+    assert torch.allclose(flat, target, atol=1e-3), (
+        f"rank {rank}: max err {(flat - target).abs().max()}"
+    )
+    # working weights == bf16 image of the master
+    assert torch.equal(
+        gdp.flatp.shadow,
+        gdp.flatp.flat[: gdp.flatp.n_cast].to(torch.bfloat16),
+    )
+    # state dict carries the fp32 master and restores exactly
+    sd = gdp.state_dict()
+    assert "master_flat" in sd
+    gdp2 = GossipDataParallel(
+        tiny_model(seed=rank + 7), push_sum=True,
+        working_dtype=torch.bfloat16,
+    )
+    gdp2.load_state_dict(sd)
+    assert torch.equal(gdp2.flatp.flat, flat)
+    gdp.shutdown()
+    gdp2.shutdown()
+
+
+def test_master_weight_consensus():
+    run_dist(_master_weight_consensus, world_size=2)

@@ -68,3 +68,108 @@ def test_scale_flat_equals_per_param_scale():
     fp.flat.mul_(0.5)
     for p, q in zip(model.parameters(), ref.parameters()):
         assert torch.allclose(p, q * 0.5)
+
+
+def test_flatparams_working_dtype_layout():
+    """bf16 working weights: matrix params re-pointed at the bf16 shadow
+    (cast section first in the master), BN-style vectors stay fp32."""
+    import torch.nn as nn
+
+    from stochastic_gradient_push_amd.ops.flat import FlatParams
+
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Linear(8, 16), nn.BatchNorm1d(16), nn.Linear(16, 4))
+    ref = [p.detach().clone() for p in m.parameters() if p.requires_grad]
+    fp = FlatParams(m, flatten_grads=True, working_dtype=torch.bfloat16)
+
+    assert fp.flat.dtype == torch.float32
+    assert fp.shadow is not None and fp.shadow.dtype == torch.bfloat16
+    n_cast_expect = 8 * 16 + 16 * 4
+    assert fp.n_cast == n_cast_expect
+    # matrix params are bf16 shadow views; vectors fp32 master views
+    for p in m.parameters():
+        if p.ndim >= 2:
+            assert p.dtype == torch.bfloat16
+            assert p.grad.dtype == torch.bfloat16
+        else:
+            assert p.dtype == torch.float32
+            assert p.grad.dtype == torch.float32
+    # master holds the original fp32 values (cast-section reordered)
+    flat_ref = torch.cat(
+        [r.reshape(-1) for r in ref if r.ndim >= 2]
+        + [r.reshape(-1) for r in ref if r.ndim < 2]
+    )
+    assert torch.equal(fp.flat, flat_ref)
+    # shadow is the bf16 image of the master's cast section
+    assert torch.equal(fp.shadow, fp.flat[:fp.n_cast].to(torch.bfloat16))
+
+    # sync_shadow propagates master edits
+    with torch.no_grad():
+        fp.flat.mul_(2.0)
+    fp.sync_shadow()
+    assert torch.equal(fp.shadow, fp.flat[:fp.n_cast].to(torch.bfloat16))
+
+    assert fp.grads_wired()
+    for p in m.parameters():
+        p.grad = None
+    assert not fp.grads_wired()
+    fp.rewire_grads()
+    assert fp.grads_wired()
+
+
+def test_fused_sgd_mixed_matches_torch():
+    """FusedSGD in master-weight mode == torch.optim.SGD on an fp32
+    model fed the same (bf16-rounded) gradients."""
+    import torch.nn as nn
+
+    from stochastic_gradient_push_amd.ops.flat import FlatParams
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(1)
+    m = nn.Sequential(nn.Linear(6, 8), nn.BatchNorm1d(8), nn.Linear(8, 3))
+    import copy
+
+    oracle = copy.deepcopy(m)
+    fp = FlatParams(m, flatten_grads=True, working_dtype=torch.bfloat16)
+    opt = FusedSGD(fp, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    oracle_params = (
+        [p for p in oracle.parameters() if p.ndim >= 2]
+        + [p for p in oracle.parameters() if p.ndim < 2]
+    )
+    oopt = torch.optim.SGD(
+        oracle_params, lr=0.1, momentum=0.9, weight_decay=1e-4
+    )
+
+    x = torch.randn(5, 6)
+    y = torch.randn(5, 3)
+    for step in range(3):
+        opt.zero_grad()
+        with torch.autocast(device_type="cpu", dtype=torch.bfloat16):
+            loss = ((m(x) - y) ** 2).mean()
+        loss.backward()
+        # oracle gets the SAME grads the mixed optimizer sees
+        with torch.no_grad():
+            offset = 0
+            for p in oracle_params:
+                n = p.numel()
+                if p.ndim >= 2:
+                    p.grad = fp.flat_grad_w.narrow(0, offset, n).view_as(
+                        p
+                    ).float()
+                else:
+                    p.grad = fp.flat_grad.narrow(
+                        0, offset - fp.n_cast, n
+                    ).view_as(p).clone()
+                offset += n
+        opt.step()
+        oopt.step()
+        oracle_flat = torch.cat(
+            [p.detach().reshape(-1) for p in oracle_params]
+        )
+        assert torch.allclose(fp.flat, oracle_flat, atol=1e-5), (
+            step, (fp.flat - oracle_flat).abs().max()
+        )
+        # working weights track the master
+        assert torch.equal(
+            fp.shadow, fp.flat[:fp.n_cast].to(torch.bfloat16)
+        )
