@@ -297,3 +297,65 @@ def _master_weight_consensus(rank, world_size):
 
 def test_master_weight_consensus():
     run_dist(_master_weight_consensus, world_size=2)
+
+
+def _osgp_graphed_order(rank, world_size):
+    """bench.py's hipGraph-mode OSGP step order (merge previous round,
+    kick the next exchange, THEN run the captured compute;
+    bench.py step()) must preserve push-sum mass and reach consensus —
+    the N>1 numerics the graph path relies on, emulated on CPU/gloo
+    with the same lazy-mixing state machine."""
+    from stochastic_gradient_push_amd import GossipDataParallel
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    model = tiny_model(seed=rank)
+    flat0 = torch.cat([
+        p.detach().reshape(-1) for p in model.parameters() if p.ndim >= 1
+    ])
+    gdp = GossipDataParallel(model, push_sum=True, overlap=False)
+    assert gdp.lazy_mixing  # the precondition bench's graph mode needs
+    target = torch.empty_like(gdp.flatp.flat)
+    target.copy_(gdp.flatp.flat)
+    dist.all_reduce(target)
+    target /= world_size
+
+    opt = FusedSGD(gdp.flatp, lr=0.0)
+    loss_fn = nn.CrossEntropyLoss()
+    x = torch.randn(2, 3, 8, 8)
+    y = torch.randint(0, 10, (2,))
+    gdp.train()
+
+    def compute_step():
+        # stands in for graph.replay(): fwd/bwd/optimizer only, hooks
+        # suppressed exactly as in capture (bench captures the inner
+        # module, not the wrapper)
+        loss = loss_fn(gdp.module(x), y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+
+    total0 = gdp.flatp.flat.clone()
+    dist.all_reduce(total0)
+    for it in range(40):
+        gdp._query_gossip_queue(non_blocking=gdp.asynch)
+        gdp.transfer_params()
+        compute_step()
+        # push-sum invariant: global de-biased mass is conserved
+        # (ps_weight-weighted sum of numerators == sum of params)
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    gdp.unbias()
+    flat = gdp.flatp.flat.detach()
+    assert torch.allclose(flat, target, atol=1e-3), (
+        f"rank {rank}: max err {(flat - target).abs().max()}"
+    )
+    gdp.shutdown()
+    del flat0
+
+
+def test_osgp_graphed_step_order_consensus():
+    run_dist(_osgp_graphed_order, world_size=2)
+
+
+def test_osgp_graphed_step_order_consensus_w4():
+    run_dist(_osgp_graphed_order, world_size=4)
