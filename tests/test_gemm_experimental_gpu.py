@@ -1,15 +1,12 @@
-"""EXPERIMENTAL kernels pending hardware validation (round 2).
-
-Run with: python -m pytest tests -m gpu_experimental
-(deliberately NOT part of the `gpu` tier until validated once.)
-"""
+"""wgrad TN kernel numerics (hardware-validated in round 2; promoted
+from the gpu_experimental tier to the standard gpu tier)."""
 
 import pytest
 import torch
 
 from stochastic_gradient_push_amd import ops
 
-pytestmark = pytest.mark.gpu_experimental
+pytestmark = pytest.mark.gpu
 
 
 def dev():
