@@ -48,9 +48,11 @@ def parse_args():
         help="disable hipGraph capture of the local train step",
     )
     p.add_argument(
-        "--conv-impl", type=str, default="miopen",
-        choices=["miopen", "gemm", "mfma"],
-        help="1x1 convs: MIOpen conv kernels vs hipBLASLt GEMM dispatch",
+        "--conv-impl", type=str, default="auto",
+        choices=["auto", "miopen", "gemm", "mfma"],
+        help="conv backend: auto = hand-written MFMA kernels on the "
+             "shapes where they beat MIOpen (measured per-shape table), "
+             "MIOpen elsewhere; mfma = hand-written everywhere possible",
     )
     p.add_argument(
         "--opt", type=str, default="fused",

@@ -111,6 +111,10 @@ def build_parser():
     p.add_argument("--model", default="resnet50", type=str)
     p.add_argument("--norm", default="fused", type=str,
                    choices=["fused", "native", "miopen"])
+    p.add_argument("--conv_impl", default="auto", type=str,
+                   choices=["auto", "miopen", "gemm", "mfma"],
+                   help="conv backend (auto: hand-written MFMA kernels on "
+                        "the measured winning shapes, MIOpen elsewhere)")
     p.add_argument("--num_classes", default=1000, type=int)
     p.add_argument("--fused_sgd", default="True", type=str,
                    help="use the one-kernel flat-buffer SGD")
@@ -554,6 +558,7 @@ def init_model(args):
     model = build_resnet(
         args.model, num_classes=args.num_classes,
         zero_init_residual=True, norm=args.norm,
+        conv_impl=args.conv_impl,
     )
     model = model.to(args.device)
     if args.device == "cuda":

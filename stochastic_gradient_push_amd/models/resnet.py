@@ -18,9 +18,28 @@ import torch
 import torch.nn as nn
 
 
+# Measured per-shape winners at bs=32/224px on MI355X, fwd+dgrad+wgrad
+# summed (profiles/r02_conv_bench2.txt; tools/conv_bench.py): shapes
+# where the hand-written MFMA kernels beat MIOpen end to end.  Keys are
+# (cin, cout, stride); ResNet-50's channel pairs identify its shapes
+# uniquely.  conv_impl='auto' uses MFMA on winners, MIOpen elsewhere.
+_MFMA_3X3_WINNERS = {
+    (64, 64, 1), (128, 128, 1), (256, 256, 1),
+    (512, 512, 2), (512, 512, 1),
+}
+_MFMA_1X1_WINNERS = {
+    (64, 256, 1), (256, 128, 1), (256, 512, 2), (128, 512, 1),
+    (512, 128, 1), (512, 256, 1), (512, 1024, 2), (256, 1024, 1),
+    (1024, 256, 1), (1024, 512, 1),
+}
+
+
 def conv3x3(cin: int, cout: int, stride: int = 1,
             impl: str = "miopen") -> nn.Conv2d:
-    if impl == "mfma" and cin % 64 == 0:
+    use_mfma = impl == "mfma" or (
+        impl == "auto" and (cin, cout, stride) in _MFMA_3X3_WINNERS
+    )
+    if use_mfma and cin % 64 == 0:
         from .layers import MfmaConv3x3
 
         return MfmaConv3x3(cin, cout, stride=stride)
@@ -33,7 +52,9 @@ def conv1x1(cin: int, cout: int, stride: int = 1,
         from .layers import GemmConv1x1
 
         return GemmConv1x1(cin, cout, stride=stride)
-    if impl == "mfma":
+    if impl == "mfma" or (
+        impl == "auto" and (cin, cout, stride) in _MFMA_1X1_WINNERS
+    ):
         from .layers import MfmaConv1x1
 
         return MfmaConv1x1(cin, cout, stride=stride)

@@ -282,8 +282,12 @@ def test_sgd_step_bf16gs_matches_fp32():
             first_step=first,
         )
     torch.cuda.synchronize()
-    assert torch.equal(p1, p2)
-    assert torch.equal(buf1, buf2)
+    # the two kernels may contract mul+add into FMA differently ->
+    # compare to fp32 ulp, not bitwise
+    assert torch.allclose(p1, p2, atol=1e-6, rtol=1e-6), (
+        (p1 - p2).abs().max().item()
+    )
+    assert torch.allclose(buf1, buf2, atol=1e-6, rtol=1e-6)
     assert torch.equal(shadow, p1.to(torch.bfloat16))
 
 
