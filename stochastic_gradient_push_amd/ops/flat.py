@@ -18,6 +18,24 @@ from typing import List, Optional
 import torch
 
 
+def _format_view(flat: torch.Tensor, offset: int,
+                 like: torch.Tensor) -> torch.Tensor:
+    """A view of ``flat[offset:offset+numel]`` shaped like ``like`` and
+    PRESERVING its memory format: a channels_last 4-D tensor gets a
+    channels_last-strided view (elements stored NHWC inside the flat
+    buffer), so re-pointing conv weights at the flat buffer never
+    silently converts them to NCHW (MIOpen and the hand-written NHWC
+    kernels both want channels_last weights)."""
+    n = like.numel()
+    sec = flat.narrow(0, offset, n)
+    if like.dim() == 4 and like.is_contiguous(
+        memory_format=torch.channels_last
+    ) and not like.is_contiguous():
+        N, C, H, W = like.shape
+        return sec.view(N, H, W, C).permute(0, 3, 1, 2)
+    return sec.view_as(like)
+
+
 class FlatBuffer:
     """Flatten ``tensors`` (same dtype/device) into one contiguous buffer.
 
@@ -37,7 +55,7 @@ class FlatBuffer:
         offset = 0
         self.views: List[torch.Tensor] = []
         for t, n in zip(tensors, self.numels):
-            view = self.flat.narrow(0, offset, n).view_as(t)
+            view = _format_view(self.flat, offset, t)
             view.copy_(t.detach())
             self.views.append(view)
             offset += n
@@ -110,13 +128,12 @@ class FlatParams:
             )
             self.shadow.copy_(self.flat.narrow(0, 0, self.n_cast))
         # re-point parameter storages: cast set at shadow views, the
-        # rest at fp32 master views
+        # rest at fp32 master views (format-preserving, like the master
+        # views themselves)
         offset = 0
         for p, v in zip(params, self._buf.views):
             if self.shadow is not None and offset < self.n_cast:
-                p.data = self.shadow.narrow(
-                    0, offset, p.numel()
-                ).view_as(p)
+                p.data = _format_view(self.shadow, offset, p.data)
             else:
                 p.data = v
             offset += p.numel()
@@ -182,7 +199,11 @@ class FlatParams:
         offset = 0
         for p in self.params:
             n = p.numel()
-            g = self._grad_view(offset, n).view_as(p)
+            if self.shadow is not None and offset < self.n_cast:
+                g = _format_view(self.flat_grad_w, offset, p.data)
+            else:
+                base = offset - self.n_cast
+                g = _format_view(self.flat_grad, base, p.data)
             if p.grad is None:
                 g.detach().zero_()
                 p.grad = g
