@@ -66,6 +66,12 @@ def parse_args():
         help="wire format of gossip messages (bf16 halves xGMI bytes)",
     )
     p.add_argument(
+        "--steal-grads", type=str, default="auto",
+        choices=["auto", "on", "off"],
+        help="steal-mode grads + one fused gather kernel instead of "
+             "~161 per-param accumulate adds (auto = on for fused on GPU)",
+    )
+    p.add_argument(
         "--master-weights", type=str, default="auto",
         choices=["auto", "on", "off"],
         help="bf16 working weights with fp32 master (removes per-layer "
@@ -149,8 +155,12 @@ def main():
         )
         model = gdp
         if args.opt == "fused":
+            steal = args.steal_grads == "on" or (
+                args.steal_grads == "auto" and cuda
+            )
             opt = FusedSGD(
                 gdp.flatp, lr=0.1, momentum=0.9, weight_decay=1e-4,
+                steal_grads=steal,
             )
         else:
             opt = torch.optim.SGD(

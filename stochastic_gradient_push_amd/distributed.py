@@ -728,6 +728,8 @@ class GossipDataParallel(Module):
                 # intra-node grad averaging over the local group; one
                 # all-reduce when grads are flat (reference
                 # distributed.py:551-562 flattened per dtype each step)
+                if getattr(self.flatp, "steal_mode", False):
+                    self.flatp.gather_grads()
                 if self._flatten_grads and self.flatp.flat_grad is not None:
                     bufs = [self.flatp.flat_grad]
                     if self.flatp.flat_grad_w is not None:
@@ -760,7 +762,9 @@ class GossipDataParallel(Module):
 
     def __make_forward_pre_hook(self):
         def hook(*unused):
-            if self._flatten_grads:
+            if self._flatten_grads and not getattr(
+                self.flatp, "steal_mode", False
+            ):
                 # re-wire grads if an optimizer's zero_grad(set_to_none)
                 # detached them from the flat buffer
                 if not self.flatp.grads_wired():

@@ -300,3 +300,10 @@ def sgd_step_bf16gs_(
         nesterov=nesterov, first_step=first_step,
     )
     shadow.copy_(params)
+
+
+def gather_multi_(ptr_table: torch.Tensor, offsets: torch.Tensor,
+                  out: torch.Tensor) -> None:
+    """Gather scattered grad tensors (device pointer table) into the
+    flat buffer in one launch (steal-mode grads)."""
+    _ext_for(out).gather_multi_(ptr_table, offsets, out)

@@ -49,6 +49,12 @@ void sgp_sgd_step_bf16gs(float* p, const unsigned short* g, float* buf,
                          bool first, int64_t n, hipStream_t stream);
 void sgp_cast_shadow(const float* p, unsigned short* shadow, int64_t n,
                      hipStream_t stream);
+void sgp_gather_multi_f32(const float* const* srcs, const int64_t* offsets,
+                          float* out, int nparams, int64_t total,
+                          hipStream_t stream);
+void sgp_gather_multi_bf16(const unsigned short* const* srcs,
+                           const int64_t* offsets, unsigned short* out,
+                           int nparams, int64_t total, hipStream_t stream);
 void sgp_sgd_step(float* p, const float* g, float* buf, const float* lr_ptr,
                   double mu, double wd, double damp, bool nesterov,
                   bool first, int64_t n, hipStream_t stream);
@@ -166,6 +172,34 @@ void sgd_step_bf16gs_(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
       reinterpret_cast<unsigned short*>(shadow.data_ptr()),
       lr.data_ptr<float>(), momentum, weight_decay, dampening, nesterov,
       first_step, p.numel(), current_stream(p));
+}
+
+// ptr_table: int64 DEVICE tensor of source addresses (0 = missing ->
+// zero-fill); offsets: int64 DEVICE tensor [nparams+1] of flat starts.
+void gather_multi_(torch::Tensor ptr_table, torch::Tensor offsets,
+                   torch::Tensor out) {
+  TORCH_CHECK(ptr_table.is_cuda() && ptr_table.is_contiguous()
+              && ptr_table.scalar_type() == torch::kInt64);
+  TORCH_CHECK(offsets.is_cuda() && offsets.is_contiguous()
+              && offsets.scalar_type() == torch::kInt64);
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous());
+  const int nparams = (int)ptr_table.numel();
+  TORCH_CHECK(offsets.numel() == nparams + 1);
+  if (out.scalar_type() == torch::kFloat32) {
+    sgp_gather_multi_f32(
+        reinterpret_cast<const float* const*>(ptr_table.data_ptr()),
+        offsets.data_ptr<int64_t>(), out.data_ptr<float>(), nparams,
+        out.numel(), current_stream(out));
+  } else if (out.scalar_type() == torch::kBFloat16) {
+    sgp_gather_multi_bf16(
+        reinterpret_cast<const unsigned short* const*>(
+            ptr_table.data_ptr()),
+        offsets.data_ptr<int64_t>(),
+        reinterpret_cast<unsigned short*>(out.data_ptr()), nparams,
+        out.numel(), current_stream(out));
+  } else {
+    TORCH_CHECK(false, "gather_multi_: fp32 or bf16 only");
+  }
 }
 
 void cast_shadow_(torch::Tensor p, torch::Tensor shadow) {
@@ -542,6 +576,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_bf16gs_", &sgd_step_bf16gs_,
         "fused momentum-SGD step, bf16 grads + bf16 shadow write-back");
   m.def("cast_shadow_", &cast_shadow_, "fp32 master -> bf16 shadow");
+  m.def("gather_multi_", &gather_multi_,
+        "gather scattered grad tensors into the flat buffer (one launch)");
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 MFMA layout probe");
   m.def("gemm_nt_bf16", &gemm_nt_bf16,
         "C[M,N] = A[M,K] @ B[N,K]^T, bf16 MFMA, fp32 accumulate");

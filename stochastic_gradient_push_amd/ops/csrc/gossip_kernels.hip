@@ -300,6 +300,35 @@ __global__ void k_cast_shadow(const float* __restrict__ p,
     shadow[j] = f2b_(p[j]);
 }
 
+// Multi-tensor gather: concatenate nparams scattered gradient tensors
+// (device pointer table) into the flat gradient buffer in ONE launch.
+// Replaces autograd's per-parameter accumulate adds (~161 tiny kernels
+// per step) under steal-mode grads: autograd ASSIGNS fresh tensors
+// (no kernel), this gathers them.  offsets[i] is the flat start of
+// param i, offsets[nparams] = total.  Each thread binary-searches its
+// flat index into a param (log2(161) ~ 8 steps, amortized over a
+// grid-stride loop).  Templated on element type (bf16 shadow-grad
+// section vs fp32).
+template <typename T>
+__global__ void k_gather_multi(const T* const* __restrict__ srcs,
+                               const int64_t* __restrict__ offsets,
+                               T* __restrict__ out, int nparams,
+                               int64_t total) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    // binary search: largest p with offsets[p] <= i
+    int lo = 0, hi = nparams - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (offsets[mid] <= i) lo = mid;
+      else hi = mid - 1;
+    }
+    const T* src = srcs[lo];
+    out[i] = src ? src[i - offsets[lo]] : (T)0;
+  }
+}
+
 inline int grid_for(int64_t work) {
   int64_t blocks = (work + THREADS - 1) / THREADS;
   if (blocks < 1) blocks = 1;
@@ -408,6 +437,22 @@ void sgp_cast_shadow(const float* p, unsigned short* shadow, int64_t n,
   const int64_t n4 = n / 4;
   hipLaunchKernelGGL(k_cast_shadow, dim3(grid_for(n4 ? n4 : n)),
                      dim3(THREADS), 0, stream, p, shadow, n4, n);
+}
+
+void sgp_gather_multi_f32(const float* const* srcs, const int64_t* offsets,
+                          float* out, int nparams, int64_t total,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(k_gather_multi<float>,
+                     dim3(grid_for((total + 3) / 4)), dim3(THREADS), 0,
+                     stream, srcs, offsets, out, nparams, total);
+}
+
+void sgp_gather_multi_bf16(const unsigned short* const* srcs,
+                           const int64_t* offsets, unsigned short* out,
+                           int nparams, int64_t total, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gather_multi<unsigned short>,
+                     dim3(grid_for((total + 3) / 4)), dim3(THREADS), 0,
+                     stream, srcs, offsets, out, nparams, total);
 }
 
 }  // extern "C"

@@ -30,6 +30,7 @@ class FusedSGD:
         weight_decay: float = 0.0,
         dampening: float = 0.0,
         nesterov: bool = False,
+        steal_grads: bool = False,
     ):
         if nesterov and (momentum <= 0 or dampening != 0):
             raise ValueError(
@@ -40,6 +41,11 @@ class FusedSGD:
         )
         self.flatp = flatp
         self._mixed = flatp.shadow is not None
+        self._steal = steal_grads
+        if steal_grads:
+            # autograd assigns fresh grad tensors (no per-param
+            # accumulate adds); step() gathers them in one kernel
+            flatp.enable_steal_mode()
         if momentum != 0.0:
             self.momentum_buf = torch.zeros_like(flatp.flat)
         elif self._mixed:
@@ -68,6 +74,11 @@ class FusedSGD:
         ]
 
     def zero_grad(self, set_to_none: bool = False):
+        if self._steal:
+            # drop the scattered tensors; nothing to zero (the gather
+            # overwrites the whole flat buffer next step)
+            self.flatp.release_grads()
+            return
         # flat buffers are zeroed, never detached — views stay wired
         self.flatp.zero_grad()
 
@@ -89,8 +100,10 @@ class FusedSGD:
     @torch.no_grad()
     def step(self, closure=None):
         g = self.param_groups[0]
+        if self._steal:
+            self.flatp.gather_grads()
         # autograd may have detached grads (e.g. someone else's zero_grad)
-        if not self.flatp.grads_wired():
+        elif not self.flatp.grads_wired():
             self.flatp.rewire_grads()
         self.sync_lr()
         lr = self._lr_dev if self._lr_dev is not None else g["lr"]

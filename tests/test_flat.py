@@ -1,6 +1,7 @@
 """FlatParams / FlatBuffer: param re-pointing, grad accumulation into the
 flat buffer, optimizer interplay."""
 
+import pytest
 import torch
 import torch.nn as nn
 
@@ -172,4 +173,42 @@ def test_fused_sgd_mixed_matches_torch():
         # working weights track the master
         assert torch.equal(
             fp.shadow, fp.flat[:fp.n_cast].to(torch.bfloat16)
+        )
+
+
+@pytest.mark.parametrize("working", [None, torch.bfloat16])
+def test_steal_mode_matches_wired(working):
+    """FusedSGD(steal_grads=True) (assign + one fused gather) must be
+    numerically identical to the wired-views accumulate path."""
+    import copy
+
+    import torch.nn as nn
+
+    from stochastic_gradient_push_amd.ops.flat import FlatParams
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(3)
+    m1 = nn.Sequential(nn.Linear(6, 8), nn.BatchNorm1d(8), nn.Linear(8, 3))
+    m2 = copy.deepcopy(m1)
+    fp1 = FlatParams(m1, flatten_grads=True, working_dtype=working)
+    fp2 = FlatParams(m2, flatten_grads=True, working_dtype=working)
+    o1 = FusedSGD(fp1, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    o2 = FusedSGD(fp2, lr=0.1, momentum=0.9, weight_decay=1e-4,
+                  steal_grads=True)
+
+    x = torch.randn(5, 6)
+    y = torch.randn(5, 3)
+    for _ in range(3):
+        for m, o in ((m1, o1), (m2, o2)):
+            o.zero_grad()
+            if working is not None:
+                with torch.autocast(device_type="cpu",
+                                    dtype=torch.bfloat16):
+                    loss = ((m(x) - y) ** 2).mean()
+            else:
+                loss = ((m(x) - y) ** 2).mean()
+            loss.backward()
+            o.step()
+        assert torch.allclose(fp1.flat, fp2.flat, atol=1e-6), (
+            (fp1.flat - fp2.flat).abs().max()
         )
