@@ -48,7 +48,7 @@ def parse_args():
         help="disable hipGraph capture of the local train step",
     )
     p.add_argument(
-        "--conv-impl", type=str, default="auto",
+        "--conv-impl", type=str, default="miopen",
         choices=["auto", "miopen", "gemm", "mfma"],
         help="conv backend: auto = hand-written MFMA kernels on the "
              "shapes where they beat MIOpen (measured per-shape table), "

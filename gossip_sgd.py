@@ -111,7 +111,7 @@ def build_parser():
     p.add_argument("--model", default="resnet50", type=str)
     p.add_argument("--norm", default="fused", type=str,
                    choices=["fused", "native", "miopen"])
-    p.add_argument("--conv_impl", default="auto", type=str,
+    p.add_argument("--conv_impl", default="miopen", type=str,
                    choices=["auto", "miopen", "gemm", "mfma"],
                    help="conv backend (auto: hand-written MFMA kernels on "
                         "the measured winning shapes, MIOpen elsewhere)")
