@@ -212,7 +212,7 @@ class FlatParams:
 
         for grp in self._steal_groups:
             params, out = grp["params"], grp["out"]
-            fixups = []
+            fixups = set()
             ptrs = []
             for p in params:
                 g = p.grad
@@ -223,7 +223,7 @@ class FlatParams:
                 else:
                     # unexpected layout: zero-fill + per-param copy
                     ptrs.append(0)
-                    fixups.append(p)
+                    fixups.add(id(p))
             if ptrs != grp["ptrs"]:
                 grp["host"].copy_(torch.tensor(ptrs, dtype=torch.int64))
                 grp["table"].copy_(grp["host"], non_blocking=True)
@@ -245,7 +245,7 @@ class FlatParams:
                 offs = 0
                 for p in params:
                     n = p.numel()
-                    if p in fixups:
+                    if id(p) in fixups:
                         _format_view(out, offs, p.data).copy_(
                             p.grad.detach()
                         )
