@@ -195,7 +195,10 @@ class _MfmaConv3x3Fn(torch.autograd.Function):
         stride = ctx.c3_stride
         dy = dy.contiguous(memory_format=torch.channels_last)
         k = _o._ext_for(x)
-        w_nchw = w_bf.permute(0, 3, 1, 2)  # [Co,Ci,3,3] (strided view ok)
+        # MIOpen fallbacks need a CONTIGUOUS NCHW weight: a permuted
+        # strided view makes conv2d_input/conv2d_weight fall back to the
+        # naive double-precision kernels (measured 30 ms per call)
+        w_nchw = w_bf.permute(0, 3, 1, 2).contiguous()
         if stride == 1:
             w_rot = w_bf.flip(1, 2).permute(3, 1, 2, 0).contiguous()
             dx = torch.empty_like(x)
@@ -206,7 +209,7 @@ class _MfmaConv3x3Fn(torch.autograd.Function):
             )
         dw = torch.nn.grad.conv2d_weight(
             x, list(w_nchw.shape), dy, stride=stride, padding=1,
-        ).to(ctx.c3_wdtype).permute(0, 1, 2, 3)
+        ).to(ctx.c3_wdtype).contiguous(memory_format=torch.channels_last)
         return dx, dw, None
 
 
@@ -252,7 +255,7 @@ class _FusedBNFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
-                momentum, eps, training, relu):
+                momentum, eps, training, relu, counter=None):
         from .. import ops as _o
 
         k = _o._ext_for(x)
@@ -267,11 +270,19 @@ class _FusedBNFunction(torch.autograd.Function):
             )
             smean = torch.empty(C, device=x.device, dtype=torch.float32)
             sinvstd = torch.empty(C, device=x.device, dtype=torch.float32)
-            k.bn_fwd_reduce(x, scratch, M, C)
-            k.bn_fwd_finalize(
-                scratch, gamma, beta, running_mean, running_var, smean,
-                sinvstd, scale_shift, momentum, eps, M, C, True,
-            )
+            if counter is not None:
+                # reduce + last-block finalize in ONE launch
+                k.bn_fwd_reduce_finalize(
+                    x, scratch, counter, gamma, beta, running_mean,
+                    running_var, smean, sinvstd, scale_shift, momentum,
+                    eps, M, C, True,
+                )
+            else:
+                k.bn_fwd_reduce(x, scratch, M, C)
+                k.bn_fwd_finalize(
+                    scratch, gamma, beta, running_mean, running_var, smean,
+                    sinvstd, scale_shift, momentum, eps, M, C, True,
+                )
         else:
             k.bn_eval_prep(running_mean, running_var, gamma, beta,
                            scale_shift, eps, C)
@@ -283,6 +294,7 @@ class _FusedBNFunction(torch.autograd.Function):
         ctx.bn_relu = relu
         ctx.bn_training = training
         ctx.bn_has_res = residual is not None
+        ctx.bn_counter = counter
         return y
 
     @staticmethod
@@ -297,17 +309,26 @@ class _FusedBNFunction(torch.autograd.Function):
         scratch = torch.empty(
             k.bn_partials_numel(M, C), device=x.device, dtype=torch.float32
         )
-        k.bn_bwd_reduce(x, dy, y if relu else None, smean, sinvstd,
-                        scratch, M, C, relu)
         dgamma = torch.empty(C, device=x.device, dtype=torch.float32)
         dbeta = torch.empty(C, device=x.device, dtype=torch.float32)
         coef = torch.empty(3 * C, device=x.device, dtype=torch.float32)
-        k.bn_bwd_finalize(scratch, gamma, smean, sinvstd, dgamma, dbeta,
-                          coef, M, C, ctx.bn_training)
+        counter = ctx.bn_counter
+        if counter is not None and ctx.bn_training:
+            k.bn_bwd_reduce_finalize(
+                x, dy, y if relu else None, smean, sinvstd, scratch,
+                counter, gamma, dgamma, dbeta, coef, M, C, relu,
+                ctx.bn_training,
+            )
+        else:
+            k.bn_bwd_reduce(x, dy, y if relu else None, smean, sinvstd,
+                            scratch, M, C, relu)
+            k.bn_bwd_finalize(scratch, gamma, smean, sinvstd, dgamma,
+                              dbeta, coef, M, C, ctx.bn_training)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.bn_has_res else None
         k.bn_bwd_apply(x, dy, y if relu else None, dx, dres, coef, M, C, relu)
-        return (dx, dres, dgamma, dbeta, None, None, None, None, None, None)
+        return (dx, dres, dgamma, dbeta, None, None, None, None, None, None,
+                None)
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
@@ -325,6 +346,9 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             track_running_stats=True,
         )
         self.relu = relu
+        # last-block-finalize counter (device int32, lazily created; the
+        # fused reduce+finalize kernel resets it after each use)
+        self._bn_ctr = None
         # counter kept (checkpoint interchange with nn.BatchNorm2d) but
         # never incremented — see NativeBatchNorm2d
 
@@ -333,11 +357,15 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             residual is None
             or residual.is_contiguous(memory_format=torch.channels_last)
         ):
+            if self._bn_ctr is None or self._bn_ctr.device != x.device:
+                self._bn_ctr = torch.zeros(
+                    1, dtype=torch.int32, device=x.device
+                )
             with torch.amp.autocast(device_type="cuda", enabled=False):
                 return _FusedBNFunction.apply(
                     x, residual, self.weight, self.bias, self.running_mean,
                     self.running_var, self.momentum, self.eps,
-                    self.training, self.relu,
+                    self.training, self.relu, self._bn_ctr,
                 )
         # reference fallback path (also the CPU numerics oracle)
         y = torch.batch_norm(

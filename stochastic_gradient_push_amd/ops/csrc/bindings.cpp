@@ -12,6 +12,20 @@ extern "C" {
 int bn_reduce_nblocks(int64_t M, int C);
 void bn_fwd_reduce(const ushort_t* x, float* partials, int64_t M, int C,
                    hipStream_t s);
+void bn_fwd_reduce_finalize(const ushort_t* x, float* partials,
+                            unsigned int* counter, const float* gamma,
+                            const float* beta, float* rmean, float* rvar,
+                            float* smean, float* sinvstd,
+                            float* scale_shift, double momentum, double eps,
+                            int64_t M, int C, bool update_running,
+                            hipStream_t s);
+void bn_bwd_reduce_finalize(const ushort_t* x, const ushort_t* dy,
+                            const ushort_t* y, const float* smean,
+                            const float* sinvstd, float* partials,
+                            unsigned int* counter, const float* gamma,
+                            float* dgamma, float* dbeta, float* coef,
+                            int64_t M, int C, bool relu, bool training,
+                            hipStream_t s);
 void bn_fwd_finalize(const float* shadows, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
                      float* smean, float* sinvstd, float* scale_shift,
@@ -483,6 +497,59 @@ void bn_fwd_finalize_py(torch::Tensor scratch, torch::Tensor gamma,
                   current_stream(scratch));
 }
 
+void bn_fwd_reduce_finalize_py(
+    torch::Tensor x, torch::Tensor scratch, torch::Tensor counter,
+    torch::Tensor gamma, torch::Tensor beta, torch::Tensor rmean,
+    torch::Tensor rvar, torch::Tensor smean, torch::Tensor sinvstd,
+    torch::Tensor scale_shift, double momentum, double eps, int64_t M,
+    int64_t C, bool update_running) {
+  check_act(x, "x");
+  check_f32(scratch, "partials", bn_partials_numel(M, C));
+  TORCH_CHECK(counter.is_cuda() && counter.scalar_type() == torch::kInt32
+              && counter.numel() == 1, "counter must be int32[1] device");
+  check_f32(gamma, "gamma", C);
+  check_f32(beta, "beta", C);
+  check_f32(scale_shift, "scale_shift", 2 * C);
+  bn_fwd_reduce_finalize(
+      bf16_ptr(x), scratch.data_ptr<float>(),
+      reinterpret_cast<unsigned int*>(counter.data_ptr()),
+      gamma.data_ptr<float>(), beta.data_ptr<float>(),
+      rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+      smean.data_ptr<float>(), sinvstd.data_ptr<float>(),
+      scale_shift.data_ptr<float>(), momentum, eps, M, (int)C,
+      update_running, current_stream(x));
+}
+
+void bn_bwd_reduce_finalize_py(
+    torch::Tensor x, torch::Tensor dy, torch::optional<torch::Tensor> y,
+    torch::Tensor smean, torch::Tensor sinvstd, torch::Tensor scratch,
+    torch::Tensor counter, torch::Tensor gamma, torch::Tensor dgamma,
+    torch::Tensor dbeta, torch::Tensor coef, int64_t M, int64_t C,
+    bool relu, bool training) {
+  check_act(x, "x");
+  check_act(dy, "dy");
+  check_f32(scratch, "partials", bn_partials_numel(M, C));
+  TORCH_CHECK(counter.is_cuda() && counter.scalar_type() == torch::kInt32
+              && counter.numel() == 1, "counter must be int32[1] device");
+  check_f32(gamma, "gamma", C);
+  check_f32(dgamma, "dgamma", C);
+  check_f32(dbeta, "dbeta", C);
+  check_f32(coef, "coef", 3 * C);
+  const ushort_t* yp = nullptr;
+  if (relu) {
+    TORCH_CHECK(y.has_value(), "y required for relu backward");
+    check_act(y.value(), "y");
+    yp = bf16_ptr(y.value());
+  }
+  bn_bwd_reduce_finalize(
+      bf16_ptr(x), bf16_ptr(dy), yp, smean.data_ptr<float>(),
+      sinvstd.data_ptr<float>(), scratch.data_ptr<float>(),
+      reinterpret_cast<unsigned int*>(counter.data_ptr()),
+      gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+      dbeta.data_ptr<float>(), coef.data_ptr<float>(), M, (int)C, relu,
+      training, current_stream(x));
+}
+
 void bn_eval_prep_py(torch::Tensor rmean, torch::Tensor rvar,
                      torch::Tensor gamma, torch::Tensor beta,
                      torch::Tensor scale_shift, double eps, int64_t C) {
@@ -607,6 +674,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_scale_bf16_", &add_scale_bf16_,
         "x = (x + float(r_bf16)) * a (wire-format accumulate)");
   m.def("bn_partials_numel", &bn_partials_numel);
+  m.def("bn_fwd_reduce_finalize", &bn_fwd_reduce_finalize_py,
+        "fused BN fwd reduce + last-block finalize (one launch)");
+  m.def("bn_bwd_reduce_finalize", &bn_bwd_reduce_finalize_py,
+        "fused BN bwd reduce + last-block finalize (one launch)");
   m.def("bn_fwd_reduce", &bn_fwd_reduce_py);
   m.def("bn_fwd_finalize", &bn_fwd_finalize_py);
   m.def("bn_eval_prep", &bn_eval_prep_py);

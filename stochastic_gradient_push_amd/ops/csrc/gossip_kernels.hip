@@ -309,23 +309,60 @@ __global__ void k_cast_shadow(const float* __restrict__ p,
 // flat index into a param (log2(161) ~ 8 steps, amortized over a
 // grid-stride loop).  Templated on element type (bf16 shadow-grad
 // section vs fp32).
+// Each thread handles an 8-element granule: ONE search per granule
+// (offsets staged in LDS — 8 dependent global loads per element was
+// latency-bound at 154 us for the 25.6 M-element ResNet-50 gather),
+// and the common in-param granule copies as 4 x 4 B when the source is
+// 4 B co-aligned.
 template <typename T>
 __global__ void k_gather_multi(const T* const* __restrict__ srcs,
                                const int64_t* __restrict__ offsets,
                                T* __restrict__ out, int nparams,
                                int64_t total) {
+  extern __shared__ int64_t soff[];  // [nparams + 1]
+  for (int p = threadIdx.x; p <= nparams; p += blockDim.x)
+    soff[p] = offsets[p];
+  __syncthreads();
+
+  const int64_t g_total = (total + 7) >> 3;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += stride) {
-    // binary search: largest p with offsets[p] <= i
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       g < g_total; g += stride) {
+    const int64_t i0 = g << 3;
+    const int64_t i1 = min(i0 + 8, total);
     int lo = 0, hi = nparams - 1;
     while (lo < hi) {
       const int mid = (lo + hi + 1) >> 1;
-      if (offsets[mid] <= i) lo = mid;
+      if (soff[mid] <= i0) lo = mid;
       else hi = mid - 1;
     }
     const T* src = srcs[lo];
-    out[i] = src ? src[i - offsets[lo]] : (T)0;
+    const int64_t off = soff[lo];
+    const int64_t end = soff[lo + 1];
+    if (i1 <= end) {
+      // whole granule inside one param
+      if (src == nullptr) {
+        for (int64_t i = i0; i < i1; ++i) out[i] = (T)0;
+      } else if (i1 - i0 == 8 && (((i0 - off) * sizeof(T)) & 3) == 0 &&
+                 ((uintptr_t)(src + (i0 - off)) & 3) == 0) {
+        const unsigned int* s =
+            reinterpret_cast<const unsigned int*>(src + (i0 - off));
+        unsigned int* d = reinterpret_cast<unsigned int*>(out + i0);
+        const int words = 8 * sizeof(T) / 4;
+#pragma unroll
+        for (int wdx = 0; wdx < words; ++wdx) d[wdx] = s[wdx];
+      } else {
+        for (int64_t i = i0; i < i1; ++i) out[i] = src[i - off];
+      }
+    } else {
+      // granule straddles a param boundary: per-element path
+      for (int64_t i = i0; i < i1; ++i) {
+        int l2 = lo;
+        while (l2 + 1 <= nparams - 1 && soff[l2 + 1] <= i) ++l2;
+        const T* s2 = srcs[l2];
+        out[i] = s2 ? s2[i - soff[l2]] : (T)0;
+      }
+    }
   }
 }
 
@@ -442,16 +479,18 @@ void sgp_cast_shadow(const float* p, unsigned short* shadow, int64_t n,
 void sgp_gather_multi_f32(const float* const* srcs, const int64_t* offsets,
                           float* out, int nparams, int64_t total,
                           hipStream_t stream) {
+  const size_t shmem = (nparams + 1) * sizeof(int64_t);
   hipLaunchKernelGGL(k_gather_multi<float>,
-                     dim3(grid_for((total + 3) / 4)), dim3(THREADS), 0,
+                     dim3(grid_for((total + 7) / 8)), dim3(THREADS), shmem,
                      stream, srcs, offsets, out, nparams, total);
 }
 
 void sgp_gather_multi_bf16(const unsigned short* const* srcs,
                            const int64_t* offsets, unsigned short* out,
                            int nparams, int64_t total, hipStream_t stream) {
+  const size_t shmem = (nparams + 1) * sizeof(int64_t);
   hipLaunchKernelGGL(k_gather_multi<unsigned short>,
-                     dim3(grid_for((total + 3) / 4)), dim3(THREADS), 0,
+                     dim3(grid_for((total + 7) / 8)), dim3(THREADS), shmem,
                      stream, srcs, offsets, out, nparams, total);
 }
 
