@@ -159,8 +159,8 @@ class _MfmaConv3x3Fn(torch.autograd.Function):
     dgrad: stride 1 -> the SAME kernel on dy with the rotated/transposed
            weight w_rot[ci,r',s',co] = w[co,2-r',2-s',ci];
            stride 2 -> torch.nn.grad.conv2d_input (MIOpen fallback)
-    wgrad: torch.nn.grad.conv2d_weight (MIOpen; implicit-TN kernel is
-           future work)
+    wgrad: hand-written implicit-TN kernel (split-M partials +
+           deterministic reduce), both strides
     """
 
     @staticmethod
@@ -207,9 +207,15 @@ class _MfmaConv3x3Fn(torch.autograd.Function):
             dx = torch.nn.grad.conv2d_input(
                 list(x.shape), w_nchw, dy, stride=stride, padding=1,
             )
-        dw = torch.nn.grad.conv2d_weight(
-            x, list(w_nchw.shape), dy, stride=stride, padding=1,
-        ).to(ctx.c3_wdtype).contiguous(memory_format=torch.channels_last)
+        co, ci = w_bf.shape[0], w_bf.shape[3]
+        dw_flat = torch.empty(
+            co * 9 * ci, device=x.device, dtype=torch.float32
+        )
+        k.conv3x3_wgrad_bf16(x, dy, dw_flat, stride)
+        # [Co*3*3*Ci] fp32 -> channels_last-strided [Co,Ci,3,3] grad
+        dw = dw_flat.view(co, 3, 3, ci).permute(0, 3, 1, 2).to(
+            ctx.c3_wdtype
+        )
         return dx, dw, None
 
 

@@ -2,6 +2,8 @@
 // c10::hip stream API directly — no CUDA-compat layer).
 
 #include <hip/hip_runtime.h>
+
+#include <algorithm>
 #include <torch/extension.h>
 
 #include <c10/hip/HIPStream.h>
@@ -96,6 +98,10 @@ void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
                            ushort_t* Y, float* P, int Nb, int H, int W,
                            int Ci, int Co, int Ho, int Wo, int stride,
                            int split, hipStream_t s);
+void sgp_conv3x3_wgrad_bf16(const ushort_t* dy, const ushort_t* X,
+                            float* partials, float* dw, int Nb, int H,
+                            int W, int Ci, int Co, int Ho, int Wo,
+                            int stride, int split, hipStream_t s);
 }
 
 namespace {
@@ -442,6 +448,44 @@ void conv3x3_nhwc_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor y,
       Ho, Wo, (int)stride, (int)split, current_stream(x));
 }
 
+// 3x3 wgrad: dW (fp32, [Co*3*3*Ci] = channels_last memory order of the
+// [Co,Ci,3,3] gradient) from channels_last x and dy.
+void conv3x3_wgrad_bf16(torch::Tensor x, torch::Tensor dy,
+                        torch::Tensor dw, int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16
+              && x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "x must be bf16 channels_last");
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16
+              && dy.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "dy must be bf16 channels_last");
+  const int Nb = (int)x.size(0), Ci = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Co = (int)dy.size(1), Ho = (int)dy.size(2),
+            Wo = (int)dy.size(3);
+  TORCH_CHECK(dy.size(0) == Nb);
+  TORCH_CHECK(Ci % 64 == 0, "Ci must be a multiple of 64");
+  TORCH_CHECK(stride == 1 || stride == 2);
+  TORCH_CHECK(dw.is_cuda() && dw.is_contiguous()
+              && dw.scalar_type() == torch::kFloat32
+              && dw.numel() == (int64_t)Co * 9 * Ci,
+              "dw must be fp32 [Co*9*Ci]");
+  const int64_t M = (int64_t)Nb * Ho * Wo;
+  const int K9 = 9 * Ci;
+  const int64_t tiles = ((Co + 127) / 128) * (int64_t)((K9 + 127) / 128);
+  int64_t split = 768 / (tiles > 0 ? tiles : 1);
+  split = std::min<int64_t>(split, 64);
+  split = std::min<int64_t>(split, std::max<int64_t>(1, M / 128));
+  const int64_t max_mem = (128ll << 20) / ((int64_t)Co * K9 * 4);
+  split = std::max<int64_t>(1, std::min(split, max_mem));
+  torch::Tensor partials = torch::empty(
+      {split * Co * (int64_t)K9}, x.options().dtype(torch::kFloat32));
+  sgp_conv3x3_wgrad_bf16(
+      reinterpret_cast<const ushort_t*>(dy.data_ptr()),
+      reinterpret_cast<const ushort_t*>(x.data_ptr()),
+      partials.data_ptr<float>(), dw.data_ptr<float>(), Nb, H, W, Ci, Co,
+      Ho, Wo, (int)stride, (int)split, current_stream(x));
+}
+
 // ---------------------------------------------------------------- BN ops
 
 const ushort_t* bf16_ptr(const torch::Tensor& t) {
@@ -666,6 +710,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_nhwc_bf16", &conv3x3_nhwc_bf16, py::arg("x"),
         py::arg("w"), py::arg("y"), py::arg("stride") = 1,
         "MFMA implicit-GEMM 3x3 conv, NHWC bf16, pad=1");
+  m.def("conv3x3_wgrad_bf16", &conv3x3_wgrad_bf16, py::arg("x"),
+        py::arg("dy"), py::arg("dw"), py::arg("stride") = 1,
+        "MFMA implicit-TN 3x3 wgrad (split-M partials + reduce)");
   m.def("gemm_tn_wgrad_bf16", &gemm_tn_wgrad_bf16,
         "EXPERIMENTAL: dW = dy^T @ x with split-M partials (round-2 "
         "validation pending)");

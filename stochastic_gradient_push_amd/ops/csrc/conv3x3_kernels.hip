@@ -242,6 +242,177 @@ inline int conv_grid(int64_t tiles) {
   return g < 8 ? 8 : g;
 }
 
+// --------------------------------------------------- 3x3 wgrad (TN)
+// dW[co, (r,s,ci)] = sum_m dy[m, co] * im2col(x)[m, (r,s,ci)] — the
+// split-M TN GEMM (gemm1x1_kernels.hip wgrad geometry: m-fastest lane
+// staging, transposed LDS images) with IMPLICIT im2col addressing on
+// the B operand: per m-row decode (n,ho,wo), per column-group derive
+// the filter tap (r,s) and channel ci0, pad-guard, one bf16x8 load.
+// Ci % 64 == 0 keeps every 8-column group inside one tap.  fp32
+// partials [split][Co][9Ci] + deterministic reduce.
+#define WLDS_STRIDE 40  // 32 m-cols + 8 pad
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8w;
+typedef __attribute__((ext_vector_type(4))) float f32x4w;
+
+template <int STRIDE>
+__global__ __launch_bounds__(256) void k_conv3x3_wgrad(
+    const ushort_t* __restrict__ dy,  // [N, Ho, Wo, Co] (flat [M, Co])
+    const ushort_t* __restrict__ X,   // [N, H, W, Ci]
+    float* __restrict__ partials,     // [split][Co][9*Ci]
+    int Nb, int H, int W, int Ci, int Co, int Ho, int Wo, int split) {
+  __shared__ ushort_t As[128 * WLDS_STRIDE];  // [co][m]
+  __shared__ ushort_t Bs[128 * WLDS_STRIDE];  // [col][m]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wave >> 1) * 64;  // co offset
+  const int wn = (wave & 1) * 64;   // col offset
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;
+
+  const int64_t M = (int64_t)Nb * Ho * Wo;
+  const int K9 = 9 * Ci;
+  const int co_tiles = (Co + 127) / 128;
+  const int col_tiles = (K9 + 127) / 128;
+  const int64_t tiles = (int64_t)co_tiles * col_tiles * split;
+
+  const int mrow = tid & 31;  // m-fastest in 16-lane groups
+  const int cgrp = tid >> 5;  // 8 col-groups; this thread also +8
+
+  const int64_t bid0 = conv_xcd_bid();
+  for (int64_t t = bid0; t < tiles; t += gridDim.x) {
+    const int s = (int)(t % split);
+    const int64_t ct = t / split;
+    const int tco = (int)(ct / col_tiles) * 128;
+    const int tcol = (int)(ct % col_tiles) * 128;
+
+    const int64_t m0 = (M * s) / split;
+    const int64_t m1 = (M * (s + 1)) / split;
+
+    // per-tile, per-half column decode (tap/ci0 fixed across k-steps)
+    int h_r[2], h_s[2], h_ci0[2];
+    bool h_ok[2];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int col = tcol + (cgrp + half * 8) * 8;
+      h_ok[half] = col < K9;
+      const int tap = h_ok[half] ? col / Ci : 0;
+      h_r[half] = tap / 3;
+      h_s[half] = tap - h_r[half] * 3;
+      h_ci0[half] = h_ok[half] ? col - tap * Ci : 0;
+    }
+    const bool a_full = tco + 128 <= Co;
+
+    f32x4w acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = (f32x4w){0.f, 0.f, 0.f, 0.f};
+
+    for (int64_t k0 = m0; k0 < m1; k0 += 32) {
+      __syncthreads();
+      const int64_t gm = k0 + mrow;
+      const bool mok = gm < m1;
+      // decode this m-row once per step (shared by both halves)
+      int xn = 0, hi0 = -2, wi0 = -2;
+      if (mok) {
+        const int64_t hw = (int64_t)Ho * Wo;
+        xn = (int)(gm / hw);
+        const int rem = (int)(gm - (int64_t)xn * hw);
+        const int ho = rem / Wo;
+        const int wo = rem - ho * Wo;
+        hi0 = ho * STRIDE - 1;
+        wi0 = wo * STRIDE - 1;
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int c8 = (cgrp + half * 8) * 8;
+        // dy -> As (co columns, contiguous)
+        {
+          bf16x8w v;
+          const int gc0 = tco + c8;
+          if (mok && a_full) {
+            v = *reinterpret_cast<const bf16x8w*>(dy + gm * Co + gc0);
+          } else {
+            ushort_t tmp[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              tmp[j] = (mok && gc0 + j < Co)
+                  ? dy[gm * Co + gc0 + j] : (ushort_t)0;
+            v = *reinterpret_cast<bf16x8w*>(tmp);
+          }
+          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            As[(c8 + j) * WLDS_STRIDE + mrow] = e[j];
+        }
+        // implicit im2col(x) -> Bs (branch-free guarded vector load)
+        {
+          const int hi = hi0 + h_r[half];
+          const int wi = wi0 + h_s[half];
+          const bool ok = mok && h_ok[half]
+              && (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
+          const ushort_t* px =
+              ok ? X + ((((int64_t)xn * H + hi) * W + wi) * Ci
+                        + h_ci0[half])
+                 : X;
+          const bf16x8w v = *reinterpret_cast<const bf16x8w*>(px);
+          const bf16x8w z = (bf16x8w){0, 0, 0, 0, 0, 0, 0, 0};
+          const bf16x8w vv = ok ? v : z;
+          const ushort_t* e = reinterpret_cast<const ushort_t*>(&vv);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            Bs[(c8 + j) * WLDS_STRIDE + mrow] = e[j];
+        }
+      }
+      __syncthreads();
+
+      bf16x8w afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const bf16x8w*>(
+            As + (wm + i * 16 + frow) * WLDS_STRIDE + fk0);
+        bfrag[i] = *reinterpret_cast<const bf16x8w*>(
+            Bs + (wn + i * 16 + frow) * WLDS_STRIDE + fk0);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+
+    float* out = partials + (int64_t)s * Co * K9;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const int gco = tco + wm + i * 16 + (lane >> 4) * 4 + rr;
+          const int gcol = tcol + wn + j * 16 + (lane & 15);
+          if (gco < Co && gcol < K9)
+            out[(int64_t)gco * K9 + gcol] = acc[i][j][rr];
+        }
+  }
+}
+
+__global__ void k_conv3x3_wgrad_reduce(const float* __restrict__ P,
+                                       float* __restrict__ dw,
+                                       int64_t numel, int split) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    float acc = 0.f;
+    for (int k = 0; k < split; ++k) acc += P[(int64_t)k * numel + i];
+    dw[i] = acc;
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -279,6 +450,29 @@ void sgp_conv3x3_nhwc_bf16(const ushort_t* X, const ushort_t* Wt,
                          dim3(256), 0, s, X, Wt, Y, nullptr, Nb, H, W, Ci,
                          Co, Ho, Wo, 1);
   }
+}
+
+void sgp_conv3x3_wgrad_bf16(const ushort_t* dy, const ushort_t* X,
+                            float* partials, float* dw, int Nb, int H,
+                            int W, int Ci, int Co, int Ho, int Wo,
+                            int stride, int split, hipStream_t s) {
+  const int K9 = 9 * Ci;
+  const int64_t tiles =
+      (int64_t)((Co + 127) / 128) * ((K9 + 127) / 128) * split;
+  int grid = (int)(tiles > 16384 ? 16384 : tiles);
+  grid = (grid + 7) & ~7;
+  if (grid < 8) grid = 8;
+  if (stride == 1)
+    hipLaunchKernelGGL(k_conv3x3_wgrad<1>, dim3(grid), dim3(256), 0, s,
+                       dy, X, partials, Nb, H, W, Ci, Co, Ho, Wo, split);
+  else
+    hipLaunchKernelGGL(k_conv3x3_wgrad<2>, dim3(grid), dim3(256), 0, s,
+                       dy, X, partials, Nb, H, W, Ci, Co, Ho, Wo, split);
+  const int64_t numel = (int64_t)Co * K9;
+  int rgrid = (int)(((numel + 255) / 256) > 8192 ? 8192
+                                                 : (numel + 255) / 256);
+  hipLaunchKernelGGL(k_conv3x3_wgrad_reduce, dim3(rgrid < 1 ? 1 : rgrid),
+                     dim3(256), 0, s, partials, dw, numel, split);
 }
 
 }  // extern "C"
