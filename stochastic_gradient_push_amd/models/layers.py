@@ -363,10 +363,11 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             residual is None
             or residual.is_contiguous(memory_format=torch.channels_last)
         ):
-            if self._bn_ctr is None or self._bn_ctr.device != x.device:
-                self._bn_ctr = torch.zeros(
-                    1, dtype=torch.int32, device=x.device
-                )
+            # NOTE: the fused reduce+finalize path (last-block
+            # counter) measured 3x SLOWER end to end: the per-block
+            # __threadfence forces a cross-XCD L2 writeback on MI355X
+            # (non-coherent per-XCD L2s), ~256 flushes per BN call.
+            # Kept implemented + tested but not enabled (counter=None).
             with torch.amp.autocast(device_type="cuda", enabled=False):
                 return _FusedBNFunction.apply(
                     x, residual, self.weight, self.bias, self.running_mean,

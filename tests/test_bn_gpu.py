@@ -247,3 +247,45 @@ def test_trainer_hip_graph_smoke():
         assert len(rows) >= 3
         loss = float(rows[-1].split(",")[11])
         assert loss == loss and loss < 100  # finite
+
+
+def test_fused_reduce_finalize_matches_separate():
+    """The one-launch reduce+finalize (last-block counter) must produce
+    identical statistics to the two-launch path.  (Correct but DISABLED
+    on the hot path: its per-block device fence costs a cross-XCD L2
+    writeback on MI355X — see layers.py note.)"""
+    from stochastic_gradient_push_amd import ops as O
+
+    k = O._ext_for(torch.empty(1, device=dev()))
+    N, C, H, W = 4, 128, 16, 16
+    x, _ = make_inputs(N, C, H, W, seed=9)
+    x = x.detach()
+    M = N * H * W
+    gamma = torch.rand(C, device=dev()) + 0.5
+    beta = torch.randn(C, device=dev())
+
+    def run(fused):
+        scratch = torch.empty(k.bn_partials_numel(M, C), device=dev())
+        rmean = torch.zeros(C, device=dev())
+        rvar = torch.ones(C, device=dev())
+        smean = torch.empty(C, device=dev())
+        sinvstd = torch.empty(C, device=dev())
+        ss = torch.empty(2 * C, device=dev())
+        if fused:
+            ctr = torch.zeros(1, dtype=torch.int32, device=dev())
+            k.bn_fwd_reduce_finalize(x, scratch, ctr, gamma, beta, rmean,
+                                     rvar, smean, sinvstd, ss, 0.1, 1e-5,
+                                     M, C, True)
+            torch.cuda.synchronize()
+            assert ctr.item() == 0  # reset for re-use
+        else:
+            k.bn_fwd_reduce(x, scratch, M, C)
+            k.bn_fwd_finalize(scratch, gamma, beta, rmean, rvar, smean,
+                              sinvstd, ss, 0.1, 1e-5, M, C, True)
+            torch.cuda.synchronize()
+        return smean, sinvstd, ss, rmean, rvar
+
+    a = run(False)
+    b = run(True)
+    for t1, t2 in zip(a, b):
+        assert torch.allclose(t1, t2, atol=1e-6), (t1 - t2).abs().max()
