@@ -261,8 +261,8 @@ __global__ __launch_bounds__(256) void k_conv3x3_wgrad(
     const ushort_t* __restrict__ X,   // [N, H, W, Ci]
     float* __restrict__ partials,     // [split][Co][9*Ci]
     int Nb, int H, int W, int Ci, int Co, int Ho, int Wo, int split) {
-  __shared__ ushort_t As[128 * WLDS_STRIDE];  // [co][m]
-  __shared__ ushort_t Bs[128 * WLDS_STRIDE];  // [col][m]
+  __shared__ ushort_t As[2][128 * WLDS_STRIDE];  // [co][m] (x2 bufs)
+  __shared__ ushort_t Bs[2][128 * WLDS_STRIDE];  // [col][m]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -312,71 +312,82 @@ __global__ __launch_bounds__(256) void k_conv3x3_wgrad(
       for (int j = 0; j < 4; ++j)
         acc[i][j] = (f32x4w){0.f, 0.f, 0.f, 0.f};
 
-    for (int64_t k0 = m0; k0 < m1; k0 += 32) {
+    // register-staged double buffering (same as the 1x1 TN wgrad):
+    // next chunk's loads issue right after the barrier
+    bf16x8w ra[2], rb[2];
+
+#define CW_LOAD(k0_)                                                       \
+  do {                                                                     \
+    const int64_t gm = (k0_) + mrow;                                       \
+    const bool mok = gm < m1;                                              \
+    int xn = 0, hi0 = -2, wi0 = -2;                                        \
+    if (mok) {                                                             \
+      const int64_t hw = (int64_t)Ho * Wo;                                 \
+      xn = (int)(gm / hw);                                                 \
+      const int rem = (int)(gm - (int64_t)xn * hw);                        \
+      const int ho = rem / Wo;                                             \
+      const int wo = rem - ho * Wo;                                        \
+      hi0 = ho * STRIDE - 1;                                               \
+      wi0 = wo * STRIDE - 1;                                               \
+    }                                                                      \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
+      {                                                                    \
+        const int gc0 = tco + c8;                                          \
+        if (mok && a_full) {                                               \
+          ra[half] = *reinterpret_cast<const bf16x8w*>(dy + gm * Co + gc0);\
+        } else {                                                           \
+          ushort_t tmp[8];                                                 \
+          _Pragma("unroll") for (int j = 0; j < 8; ++j)                    \
+            tmp[j] = (mok && gc0 + j < Co)                                 \
+                ? dy[gm * Co + gc0 + j] : (ushort_t)0;                     \
+          ra[half] = *reinterpret_cast<bf16x8w*>(tmp);                     \
+        }                                                                  \
+      }                                                                    \
+      {                                                                    \
+        const int hi = hi0 + h_r[half];                                    \
+        const int wi = wi0 + h_s[half];                                    \
+        const bool ok = mok && h_ok[half]                                  \
+            && (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;   \
+        const ushort_t* px =                                               \
+            ok ? X + ((((int64_t)xn * H + hi) * W + wi) * Ci               \
+                      + h_ci0[half])                                       \
+               : X;                                                        \
+        const bf16x8w v = *reinterpret_cast<const bf16x8w*>(px);           \
+        const bf16x8w z = (bf16x8w){0, 0, 0, 0, 0, 0, 0, 0};               \
+        rb[half] = ok ? v : z;                                             \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+#define CW_WRITE(buf)                                                      \
+  do {                                                                     \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
+      const ushort_t* ea = reinterpret_cast<const ushort_t*>(&ra[half]);   \
+      const ushort_t* eb = reinterpret_cast<const ushort_t*>(&rb[half]);   \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
+        As[buf][(c8 + j) * WLDS_STRIDE + mrow] = ea[j];                    \
+        Bs[buf][(c8 + j) * WLDS_STRIDE + mrow] = eb[j];                    \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+    const int KTm = (int)((m1 - m0 + 31) / 32);
+    CW_LOAD(m0);
+    CW_WRITE(0);
+    for (int kt = 0; kt < KTm; ++kt) {
       __syncthreads();
-      const int64_t gm = k0 + mrow;
-      const bool mok = gm < m1;
-      // decode this m-row once per step (shared by both halves)
-      int xn = 0, hi0 = -2, wi0 = -2;
-      if (mok) {
-        const int64_t hw = (int64_t)Ho * Wo;
-        xn = (int)(gm / hw);
-        const int rem = (int)(gm - (int64_t)xn * hw);
-        const int ho = rem / Wo;
-        const int wo = rem - ho * Wo;
-        hi0 = ho * STRIDE - 1;
-        wi0 = wo * STRIDE - 1;
-      }
-#pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        const int c8 = (cgrp + half * 8) * 8;
-        // dy -> As (co columns, contiguous)
-        {
-          bf16x8w v;
-          const int gc0 = tco + c8;
-          if (mok && a_full) {
-            v = *reinterpret_cast<const bf16x8w*>(dy + gm * Co + gc0);
-          } else {
-            ushort_t tmp[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              tmp[j] = (mok && gc0 + j < Co)
-                  ? dy[gm * Co + gc0 + j] : (ushort_t)0;
-            v = *reinterpret_cast<bf16x8w*>(tmp);
-          }
-          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            As[(c8 + j) * WLDS_STRIDE + mrow] = e[j];
-        }
-        // implicit im2col(x) -> Bs (branch-free guarded vector load)
-        {
-          const int hi = hi0 + h_r[half];
-          const int wi = wi0 + h_s[half];
-          const bool ok = mok && h_ok[half]
-              && (unsigned)hi < (unsigned)H && (unsigned)wi < (unsigned)W;
-          const ushort_t* px =
-              ok ? X + ((((int64_t)xn * H + hi) * W + wi) * Ci
-                        + h_ci0[half])
-                 : X;
-          const bf16x8w v = *reinterpret_cast<const bf16x8w*>(px);
-          const bf16x8w z = (bf16x8w){0, 0, 0, 0, 0, 0, 0, 0};
-          const bf16x8w vv = ok ? v : z;
-          const ushort_t* e = reinterpret_cast<const ushort_t*>(&vv);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            Bs[(c8 + j) * WLDS_STRIDE + mrow] = e[j];
-        }
-      }
-      __syncthreads();
+      const int buf = kt & 1;
+      if (kt + 1 < KTm) CW_LOAD(m0 + (int64_t)(kt + 1) * 32);
 
       bf16x8w afrag[4], bfrag[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         afrag[i] = *reinterpret_cast<const bf16x8w*>(
-            As + (wm + i * 16 + frow) * WLDS_STRIDE + fk0);
+            As[buf] + (wm + i * 16 + frow) * WLDS_STRIDE + fk0);
         bfrag[i] = *reinterpret_cast<const bf16x8w*>(
-            Bs + (wn + i * 16 + frow) * WLDS_STRIDE + fk0);
+            Bs[buf] + (wn + i * 16 + frow) * WLDS_STRIDE + fk0);
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -384,7 +395,12 @@ __global__ __launch_bounds__(256) void k_conv3x3_wgrad(
         for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+
+      __syncthreads();
+      if (kt + 1 < KTm) CW_WRITE(buf ^ 1);
     }
+#undef CW_LOAD
+#undef CW_WRITE
 
     float* out = partials + (int64_t)s * Co * K9;
 #pragma unroll

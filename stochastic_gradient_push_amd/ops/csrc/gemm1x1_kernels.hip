@@ -927,8 +927,8 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
     const ushort_t* __restrict__ x,   // [M, Ci]
     float* __restrict__ partials,     // [split][Co, Ci] (tile-major ok)
     int64_t M, int Co, int Ci, int split) {
-  __shared__ ushort_t As[BM * LDS_STRIDE];  // [co][m] transposed image
-  __shared__ ushort_t Bs[BN * LDS_STRIDE];  // [ci][m]
+  __shared__ ushort_t As[2][BM * LDS_STRIDE];  // [co][m] images (x2)
+  __shared__ ushort_t Bs[2][BN * LDS_STRIDE];  // [ci][m]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -964,63 +964,72 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
     const bool a_full = tco + BM <= Co;
     const bool b_full = tci + BN <= Ci;
 
-    for (int64_t k0 = m0; k0 < m1; k0 += BK) {
+    // register-staged double buffering (the NT v2 structure): loads
+    // for m-chunk t+1 issue right after the barrier and hide behind
+    // the MFMA block; writes land in the other LDS buffer
+    bf16x8 ra[2], rb[2];
+
+#define TN_LOAD(k0_)                                                       \
+  do {                                                                     \
+    const int64_t gm = (k0_) + mrow;                                       \
+    const bool mok = gm < m1;                                              \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
+      {                                                                    \
+        const int gc0 = tco + c8;                                          \
+        if (mok && a_full) {                                               \
+          ra[half] = *reinterpret_cast<const bf16x8*>(dy + gm * Co + gc0); \
+        } else {                                                           \
+          ushort_t tmp[8];                                                 \
+          _Pragma("unroll") for (int j = 0; j < 8; ++j)                    \
+            tmp[j] = (mok && gc0 + j < Co)                                 \
+                ? dy[gm * Co + gc0 + j] : (ushort_t)0;                     \
+          ra[half] = *reinterpret_cast<bf16x8*>(tmp);                      \
+        }                                                                  \
+      }                                                                    \
+      {                                                                    \
+        const int gc0 = tci + c8;                                          \
+        if (mok && b_full) {                                               \
+          rb[half] = *reinterpret_cast<const bf16x8*>(x + gm * Ci + gc0);  \
+        } else {                                                           \
+          ushort_t tmp[8];                                                 \
+          _Pragma("unroll") for (int j = 0; j < 8; ++j)                    \
+            tmp[j] = (mok && gc0 + j < Ci)                                 \
+                ? x[gm * Ci + gc0 + j] : (ushort_t)0;                      \
+          rb[half] = *reinterpret_cast<bf16x8*>(tmp);                      \
+        }                                                                  \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+#define TN_WRITE(buf)                                                      \
+  do {                                                                     \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
+      const ushort_t* ea = reinterpret_cast<const ushort_t*>(&ra[half]);   \
+      const ushort_t* eb = reinterpret_cast<const ushort_t*>(&rb[half]);   \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
+        As[buf][(c8 + j) * LDS_STRIDE + mrow] = ea[j];                     \
+        Bs[buf][(c8 + j) * LDS_STRIDE + mrow] = eb[j];                     \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
+    const int KTm = (int)((m1 - m0 + BK - 1) / BK);
+    TN_LOAD(m0);
+    TN_WRITE(0);
+    for (int kt = 0; kt < KTm; ++kt) {
       __syncthreads();
-      const int64_t gm = k0 + mrow;
-      const bool mok = gm < m1;
-      // 16 channel-groups of 8 per operand row: this thread covers
-      // cgrp and cgrp+8
-#pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        const int c8 = (cgrp + half * 8) * 8;
-        // dy -> As
-        {
-          bf16x8 v;
-          const int gc0 = tco + c8;
-          if (mok && a_full) {
-            v = *reinterpret_cast<const bf16x8*>(dy + gm * Co + gc0);
-          } else {
-            ushort_t tmp[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              tmp[j] = (mok && gc0 + j < Co)
-                  ? dy[gm * Co + gc0 + j] : (ushort_t)0;
-            v = *reinterpret_cast<bf16x8*>(tmp);
-          }
-          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            As[(c8 + j) * LDS_STRIDE + mrow] = e[j];
-        }
-        // x -> Bs
-        {
-          bf16x8 v;
-          const int gc0 = tci + c8;
-          if (mok && b_full) {
-            v = *reinterpret_cast<const bf16x8*>(x + gm * Ci + gc0);
-          } else {
-            ushort_t tmp[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              tmp[j] = (mok && gc0 + j < Ci)
-                  ? x[gm * Ci + gc0 + j] : (ushort_t)0;
-            v = *reinterpret_cast<bf16x8*>(tmp);
-          }
-          const ushort_t* e = reinterpret_cast<const ushort_t*>(&v);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            Bs[(c8 + j) * LDS_STRIDE + mrow] = e[j];
-        }
-      }
-      __syncthreads();
+      const int buf = kt & 1;
+      if (kt + 1 < KTm) TN_LOAD(m0 + (int64_t)(kt + 1) * BK);
 
       bf16x8 afrag[4], bfrag[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         afrag[i] = *reinterpret_cast<const bf16x8*>(
-            As + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
+            As[buf] + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
         bfrag[i] = *reinterpret_cast<const bf16x8*>(
-            Bs + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
+            Bs[buf] + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -1028,7 +1037,12 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
         for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+
+      __syncthreads();
+      if (kt + 1 < KTm) TN_WRITE(buf ^ 1);
     }
+#undef TN_LOAD
+#undef TN_WRITE
 
     float* out = partials + (int64_t)s * Co * Ci;
 #pragma unroll
