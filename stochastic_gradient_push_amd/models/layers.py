@@ -14,8 +14,15 @@ module provides:
   backward reductions in two.
 """
 
+import os
+
 import torch
 import torch.nn as nn
+
+# one-launch BN reduce+finalize (agent-scope sc1 partial stores +
+# last-block counter).  v1 (device fences) measured 3x slower; v2 is
+# gated here until measured.
+_BN_FUSE = os.environ.get("SGP_BN_FUSE", "0") == "1"
 
 
 class NativeBatchNorm2d(nn.BatchNorm2d):
@@ -363,11 +370,12 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             residual is None
             or residual.is_contiguous(memory_format=torch.channels_last)
         ):
-            # NOTE: the fused reduce+finalize path (last-block
-            # counter) measured 3x SLOWER end to end: the per-block
-            # __threadfence forces a cross-XCD L2 writeback on MI355X
-            # (non-coherent per-XCD L2s), ~256 flushes per BN call.
-            # Kept implemented + tested but not enabled (counter=None).
+            if _BN_FUSE and (
+                self._bn_ctr is None or self._bn_ctr.device != x.device
+            ):
+                self._bn_ctr = torch.zeros(
+                    1, dtype=torch.int32, device=x.device
+                )
             with torch.amp.autocast(device_type="cuda", enabled=False):
                 return _FusedBNFunction.apply(
                     x, residual, self.weight, self.bias, self.running_mean,

@@ -74,14 +74,22 @@ __device__ __forceinline__ U4 pack8(const float* f) {
   return v;
 }
 
-// Last-block detection for reduce+finalize fusion: every block fences
-// its partial stores, bumps an agent-scope counter, and exactly one
-// block (the last to arrive) runs the finalize with the counter reset
-// for the next call.  Removes one ~5 us kernel launch per BN layer per
-// direction per step (~106/step on ResNet-50).
+// Last-block detection for reduce+finalize fusion.  VERSION 2: the
+// first attempt used __threadfence() per block (device-scope release),
+// which on MI355X forces a cross-XCD L2 writeback — measured 3x step
+// regression.  Here the PARTIALS THEMSELVES are written with
+// agent-scope (system-coherent, sc1) stores, so no fence is needed:
+// the stores bypass the non-coherent per-XCD caching and the last
+// block reads them back with agent-scope loads.
+__device__ __forceinline__ void st_agent(float* p, float v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ float ld_agent(const float* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
 __device__ __forceinline__ bool bn_signal_last(unsigned int* ctr) {
-  __threadfence();
-  __syncthreads();
+  __syncthreads();  // all of this block's partial stores issued
   __shared__ bool last;
   if (threadIdx.x == 0) {
     const unsigned int prev = __hip_atomic_fetch_add(
@@ -109,8 +117,8 @@ __device__ void bn_fwd_finalize_block(
     float s = 0.f, q = 0.f;
     if (c < C) {
       for (int k = w; k < nblocks; k += 32) {
-        s += partials[(int64_t)k * 2 * C + c];
-        q += partials[(int64_t)k * 2 * C + C + c];
+        s += ld_agent(partials + (int64_t)k * 2 * C + c);
+        q += ld_agent(partials + (int64_t)k * 2 * C + C + c);
       }
     }
     red[threadIdx.x] = s;
@@ -156,8 +164,8 @@ __device__ void bn_bwd_finalize_block(
     float dg = 0.f, db = 0.f;
     if (c < C) {
       for (int k = w; k < nblocks; k += 32) {
-        dg += partials[(int64_t)k * 2 * C + c];
-        db += partials[(int64_t)k * 2 * C + C + c];
+        dg += ld_agent(partials + (int64_t)k * 2 * C + c);
+        db += ld_agent(partials + (int64_t)k * 2 * C + C + c);
       }
     }
     red[threadIdx.x] = dg;
@@ -239,10 +247,18 @@ __global__ void k_bn_fwd_reduce(const ushort_t* __restrict__ x,
       }
     }
     float* part = partials + (int64_t)blockIdx.x * 2 * C;
+    if (FUSED) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      part[c0 + j] = s[j];
-      part[C + c0 + j] = q[j];
+      for (int j = 0; j < 8; ++j) {
+        st_agent(part + c0 + j, s[j]);
+        st_agent(part + C + c0 + j, q[j]);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        part[c0 + j] = s[j];
+        part[C + c0 + j] = q[j];
+      }
     }
   }
   if (FUSED) {
@@ -421,10 +437,18 @@ __global__ void k_bn_bwd_reduce(const ushort_t* __restrict__ x,
       }
     }
     float* part = partials + (int64_t)blockIdx.x * 2 * C;
+    if (FUSED) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      part[c0 + j] = dg[j];
-      part[C + c0 + j] = db[j];
+      for (int j = 0; j < 8; ++j) {
+        st_agent(part + c0 + j, dg[j]);
+        st_agent(part + C + c0 + j, db[j]);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        part[c0 + j] = dg[j];
+        part[C + c0 + j] = db[j];
+      }
     }
   }
   if (FUSED) {
