@@ -66,6 +66,11 @@ def parse_args():
         help="wire format of gossip messages (bf16 halves xGMI bytes)",
     )
     p.add_argument(
+        "--gossip-chunks", type=int, default=1,
+        help="split each gossip message into N chunked send/recv pairs "
+             "(lets RCCL spread a single-peer exchange over xGMI links)",
+    )
+    p.add_argument(
         "--steal-grads", type=str, default="auto",
         choices=["auto", "on", "off"],
         help="steal-mode grads + one fused gather kernel instead of "
@@ -152,6 +157,7 @@ def main():
             ),
             flatten_grads=(args.opt == "fused"),
             working_dtype=torch.bfloat16 if use_master else None,
+            gossip_chunks=args.gossip_chunks,
         )
         model = gdp
         if args.opt == "fused":

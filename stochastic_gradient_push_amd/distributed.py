@@ -82,6 +82,7 @@ class GossipDataParallel(Module):
         gossip_dtype: Optional[torch.dtype] = None,
         comm_backend: str = "c10d",
         working_dtype: Optional[torch.dtype] = None,
+        gossip_chunks: int = 1,
     ):
         super().__init__()
 
@@ -168,6 +169,7 @@ class GossipDataParallel(Module):
             "world_size": world_size,
             "cpu_comm": self.__cpu_comm,
             "gossipers": {},
+            "gossip_chunks": gossip_chunks,
         }
         self.overlap = overlap
         self.synch_freq = synch_freq
@@ -637,6 +639,7 @@ class GossipDataParallel(Module):
             logger=logger,
             group=gossip_group,
             transport=dist_config.get("transport"),
+            chunks=dist_config.get("gossip_chunks", 1),
         )
         dist_config["gossipers"] = {gossip_params.dtype: gossiper}
         gossip_ps_factor.data.copy_(gossiper.mixing_weights["lo"])

@@ -78,6 +78,7 @@ class Gossiper:
         world_size: Optional[int] = None,
         group=None,
         transport=None,
+        chunks: int = 1,
     ):
         """
         :param msg: prototype message tensor (sized like the flat params)
@@ -94,6 +95,12 @@ class Gossiper:
             :class:`~stochastic_gradient_push_amd.comm.RcclTransport`;
             when given, exchanges bypass c10d and go straight through the
             C++ comm core's own RCCL communicator.
+        :param chunks: split each peer message into this many chunked
+            send/recv pairs inside the same group.  xGMI is point-to-
+            point (7 links per MI355X GPU); with peers_per_itr=1 a single
+            ~51 MB bf16 message rides however many channels RCCL assigns
+            one send — chunking gives the scheduler independent pieces
+            to spread across links.  Tune at N>1; 1 = off.
         """
         self.logger = logger
         if rank is None or world_size is None:
@@ -104,6 +111,8 @@ class Gossiper:
         self.world_size = world_size
         self.group = group
         self.transport = transport
+        assert chunks >= 1
+        self.chunks = int(chunks)
 
         assert isinstance(graph, GraphManager)
         self._graph_manager = graph
@@ -249,6 +258,14 @@ class Gossiper:
             self._recv_pool.append(self.in_msg_buffer.clone())
         return self._recv_pool[:n]
 
+    def _chunked(self, buf: torch.Tensor) -> List[torch.Tensor]:
+        """Split a flat message into ``self.chunks`` contiguous pieces
+        (both peers split identically, so chunked sends pair with
+        chunked recvs in order within the pairwise channel)."""
+        if self.chunks <= 1 or buf.numel() < self.chunks:
+            return [buf]
+        return list(torch.chunk(buf, self.chunks))
+
     def _send_buffers(self, n: int) -> List[torch.Tensor]:
         while len(self._send_pool_) < n:
             self._send_pool_.append(self.in_msg_buffer.clone())
@@ -322,13 +339,17 @@ class Gossiper:
             for i, e in enumerate(self.out_edges):
                 assert e.src == self.rank
                 buf = send_msg[i] if per_edge else send_msg
-                ops.append(
-                    dist.P2POp(dist.isend, buf, e.dest, group=self.group)
-                )
+                for piece in self._chunked(buf):
+                    ops.append(
+                        dist.P2POp(dist.isend, piece, e.dest,
+                                   group=self.group)
+                    )
             for buf, e in zip(recvs, self.in_edges):
-                ops.append(
-                    dist.P2POp(dist.irecv, buf, e.src, group=self.group)
-                )
+                for piece in self._chunked(buf):
+                    ops.append(
+                        dist.P2POp(dist.irecv, piece, e.src,
+                                   group=self.group)
+                    )
             if ops:
                 reqs = dist.batch_isend_irecv(ops)
                 for r in reqs:

@@ -320,3 +320,34 @@ def _weighted_mass_conservation(rank, world_size):
 
 def test_weighted_mixing_mass_conserved():
     run_dist(_weighted_mass_conservation, world_size=4)
+
+
+def _chunked_average(rank, world_size):
+    """Chunked gossip messages (xGMI link spreading at N>1) pair
+    correctly and converge identically."""
+    from stochastic_gradient_push_amd import gossiper as G
+    from stochastic_gradient_push_amd.graphs import (
+        NPeerDynamicDirectedExponentialGraph,
+    )
+
+    torch.manual_seed(60 + rank)
+    graph = NPeerDynamicDirectedExponentialGraph(rank, world_size)
+    gossiper = G.PushSum(
+        torch.zeros(N), graph=graph, device=torch.device("cpu"),
+        rank=rank, world_size=world_size, chunks=3,
+    )
+    x = torch.randn(N)
+    w = torch.ones(1)
+    target = x.clone()
+    dist.all_reduce(target)
+    target /= world_size
+    for _ in range(60):
+        x, w = gossiper.mix(x.clone(), w.clone(), residual=False)
+        x = x.clone()
+        w = w.clone().reshape(1)
+    est = x / w
+    assert torch.allclose(est, target, atol=1e-4), f"rank {rank}"
+
+
+def test_chunked_gossip_converges():
+    run_dist(_chunked_average, world_size=4)
