@@ -88,6 +88,8 @@ void sgp_gemm_nt_bf16_v5(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, hipStream_t s);
 void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
                          int64_t M, int N, int K, int span, hipStream_t s);
+void sgp_gemm_nt_bf16_v7(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, int span, hipStream_t s);
 void sgp_gemm_nt_splitk_bf16(const ushort_t* A, const ushort_t* B,
                              float* P, ushort_t* C, int64_t M, int N,
                              int K, int split, hipStream_t s);
@@ -315,6 +317,19 @@ void gemm_nt_bf16_v6(torch::Tensor A, torch::Tensor B, torch::Tensor C,
               && A.size(1) % 64 == 0,
               "v6 requires M %% 256 == 0, N %% 128 == 0, K %% 64 == 0");
   sgp_gemm_nt_bf16_v6(reinterpret_cast<const ushort_t*>(A.data_ptr()),
+                      reinterpret_cast<const ushort_t*>(B.data_ptr()),
+                      reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
+                      (int)B.size(0), (int)A.size(1), span ? 1 : 0,
+                      current_stream(A));
+}
+
+void gemm_nt_bf16_v7(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                     bool span) {
+  gemm_nt_check(A, B, C);
+  TORCH_CHECK(A.size(0) % 256 == 0 && B.size(0) % 128 == 0
+              && A.size(1) % 64 == 0,
+              "v7 requires M %% 256 == 0, N %% 128 == 0, K %% 64 == 0");
+  sgp_gemm_nt_bf16_v7(reinterpret_cast<const ushort_t*>(A.data_ptr()),
                       reinterpret_cast<const ushort_t*>(B.data_ptr()),
                       reinterpret_cast<ushort_t*>(C.data_ptr()), A.size(0),
                       (int)B.size(0), (int)A.size(1), span ? 1 : 0,
@@ -702,6 +717,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("C"), py::arg("span") = false,
         "NT bf16 MFMA GEMM v6 (256x128 tile, 8 waves; span=3-buf "
         "barrier-crossing glds)");
+  m.def("gemm_nt_bf16_v7", &gemm_nt_bf16_v7, py::arg("A"), py::arg("B"),
+        py::arg("C"), py::arg("span") = true,
+        "v6 with LDS-staged vectorized epilogue");
   m.def("gemm_nt_splitk_bf16", &gemm_nt_splitk_bf16, py::arg("A"),
         py::arg("B"), py::arg("C"), py::arg("split"),
         "NT bf16 MFMA GEMM, K split over block groups (small-M shapes)");

@@ -782,7 +782,7 @@ __global__ void k_gemm_nt_reduce_bf16(const float* __restrict__ P,
 #define BN6 128
 #define TILE6 (BM6 + BN6)          // 384 rows of 128 B = 48 KB / buffer
 
-template <bool SPAN>
+template <bool SPAN, bool LDSEPI>
 __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     ushort_t* __restrict__ C, int64_t M, int N, int K) {
@@ -894,16 +894,49 @@ __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
     }
 #undef GLDS6
 
+    if (!LDSEPI) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < 4; ++j)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
-          const int gc = tn + wn + j * 16 + (lane & 15);
-          C[gr * N + gc] = f2b(acc[i][j][r]);
-        }
+          for (int r = 0; r < 4; ++r) {
+            const int64_t gr = tm + wm + i * 16 + (lane >> 4) * 4 + r;
+            const int gc = tn + wn + j * 16 + (lane & 15);
+            C[gr * N + gc] = f2b(acc[i][j][r]);
+          }
+    } else {
+      // LDS-staged epilogue: the tile buffers are dead after the last
+      // k-step, so stage the 256x128 bf16 C tile in LDS (scalar
+      // conflict-light writes) and stream it out with coalesced 16-B
+      // stores (the naive epilogue is 64 scalar 2-B stores per lane).
+      ushort_t* cs = lds;  // 256*128*2 B = 64 KB of the freed buffers
+      __syncthreads();     // everyone done reading the k-tiles
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int lr = wm + i * 16 + (lane >> 4) * 4 + r;
+            const int lc = wn + j * 16 + (lane & 15);
+            cs[lr * BN6 + lc] = f2b(acc[i][j][r]);
+          }
+      __syncthreads();
+      // 256 rows x 256 B; 512 threads -> each thread 4 x 16-B pieces
+      const int t = threadIdx.x;
+#pragma unroll
+      for (int piece = 0; piece < 4; ++piece) {
+        const int idx = t + piece * 512;      // 0..2047 16-B segments
+        const int row = idx >> 3;             // 8 segments per row
+        const int c8 = (idx & 7) * 16;        // bf16 col offset x16
+        *reinterpret_cast<bf16x8*>(C + (tm + row) * N + tn + c8) =
+            *reinterpret_cast<const bf16x8*>(cs + row * BN6 + c8);
+        *reinterpret_cast<bf16x8*>(C + (tm + row) * N + tn + c8 + 8) =
+            *reinterpret_cast<const bf16x8*>(cs + row * BN6 + c8 + 8);
+      }
+      __syncthreads();  // cs is reused as the k-tile image next tile
+    }
   }
 }
 
@@ -1131,11 +1164,25 @@ void sgp_gemm_nt_bf16_v6(const ushort_t* A, const ushort_t* B, ushort_t* C,
   int grid = (int)((tiles + 7) & ~7);
   if (grid < 8) grid = 8;
   if (span)
-    hipLaunchKernelGGL(k_gemm_nt_bf16_v6<true>, dim3(grid), dim3(512), 0, s,
-                       A, B, C, M, N, K);
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v6<true, false>), dim3(grid),
+                       dim3(512), 0, s, A, B, C, M, N, K);
   else
-    hipLaunchKernelGGL(k_gemm_nt_bf16_v6<false>, dim3(grid), dim3(512), 0, s,
-                       A, B, C, M, N, K);
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v6<false, false>), dim3(grid),
+                       dim3(512), 0, s, A, B, C, M, N, K);
+}
+
+void sgp_gemm_nt_bf16_v7(const ushort_t* A, const ushort_t* B, ushort_t* C,
+                         int64_t M, int N, int K, int span, hipStream_t s) {
+  int64_t tiles = (M / BM6) * (int64_t)(N / BN6);
+  if (tiles > 16384) tiles = 16384;
+  int grid = (int)((tiles + 7) & ~7);
+  if (grid < 8) grid = 8;
+  if (span)
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v6<true, true>), dim3(grid),
+                       dim3(512), 0, s, A, B, C, M, N, K);
+  else
+    hipLaunchKernelGGL((k_gemm_nt_bf16_v6<false, true>), dim3(grid),
+                       dim3(512), 0, s, A, B, C, M, N, K);
 }
 
 void sgp_gemm_nt_bf16_v4(const ushort_t* A, const ushort_t* B, ushort_t* C,

@@ -397,3 +397,20 @@ def test_mfma_resnet_block_forward_backward():
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss).item()
+
+
+@pytest.mark.parametrize("span", [False, True])
+def test_gemm_nt_v7_matches_v1(span):
+    """v7 (LDS-staged vectorized epilogue) must be bitwise v1."""
+    M, N, K = 25088, 512, 512
+    torch.manual_seed(8)
+    A = torch.randn(M, K, device=dev()).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev()).to(torch.bfloat16)
+    C1 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    C7 = torch.zeros(M, N, device=dev(), dtype=torch.bfloat16)
+    ext().gemm_nt_bf16(A, B, C1)
+    ext().gemm_nt_bf16_v7(A, B, C7, span=span)
+    torch.cuda.synchronize()
+    assert torch.equal(C1, C7), (
+        span, (C1.float() - C7.float()).abs().max().item()
+    )

@@ -57,7 +57,7 @@ def main():
         (25088, 512, 512),         # the ladder's reference shape
     ]
 
-    cols = ("v2", "v3", "v5", "v6", "v6s", "auto", "lib")
+    cols = ("v2", "v3", "v5", "v6", "v6s", "v7", "auto", "lib")
     print(f"{'M':>8} {'N':>5} {'K':>5} | " + " ".join(
         f"{c:>6}" for c in cols) + "  (TFLOP/s)")
     for M, N, K in shapes:
@@ -76,6 +76,8 @@ def main():
                 r["v6"] = fl / timeit(lambda: ext.gemm_nt_bf16_v6(
                     A, B, C, span=False)) / 1e12
                 r["v6s"] = fl / timeit(lambda: ext.gemm_nt_bf16_v6(
+                    A, B, C, span=True)) / 1e12
+                r["v7"] = fl / timeit(lambda: ext.gemm_nt_bf16_v7(
                     A, B, C, span=True)) / 1e12
         r["auto"] = fl / timeit(lambda: ops.gemm_nt(A, B, out=C)) / 1e12
         r["lib"] = fl / timeit(lambda: torch.matmul(A, B.t())) / 1e12
