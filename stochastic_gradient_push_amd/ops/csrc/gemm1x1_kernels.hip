@@ -864,6 +864,7 @@ __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
       const int buf = SPAN ? (kt % 3) : (kt & 1);
       const ushort_t* As_ = lds + buf * TILE6 * BK3;
       const ushort_t* Bs_ = As_ + BM6 * BK3;
+      __builtin_amdgcn_s_setprio(1);  // prioritize the MFMA cluster
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         bf16x8 afrag[4], bfrag[4];
@@ -884,6 +885,7 @@ __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
       if (SPAN) {
         // passing this round's barrier proves all waves read tile kt-1:
         // its buffer (kt+2)%3 is safe to refill with no extra barrier
