@@ -194,8 +194,9 @@ def sgd_step_(
 # Variant choice per shape; see profiles/ for the measured ladder.
 
 #: use the 3-buffer barrier-crossing glds variant of the 256x128 kernel
-#: (set from hardware measurement; toggled by tools/gemm_bench.py results)
-V6_SPAN = True
+#: (measured across boxes: 2-buffer syncthreads wins or ties the span
+#: variant at this occupancy — 526/523/527 vs 501/521/531 TF)
+V6_SPAN = False
 
 
 def gemm_nt(A: torch.Tensor, B: torch.Tensor,
