@@ -250,10 +250,11 @@ def gemm_tn_wgrad(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     Ci = x.shape[1]
     co_tiles = (Co + 127) // 128
     ci_tiles = (Ci + 127) // 128
-    # enough splits to fill the chip (~768 blocks), bounded by the
-    # reduce kernel's depth (deep strided partial sums turn the reduce
-    # into the bottleneck — measured at split 2048) and the workspace
-    split = max(1, 768 // (co_tiles * ci_tiles))
+    # split sweep (profiles: r02_wgrad_split_sweep) peaks at
+    # tiles*split ~ 512 blocks (2/CU) and falls beyond (partial
+    # traffic + barrier amortization); deep strided reduces were the
+    # bottleneck at split 2048
+    split = max(1, 512 // (co_tiles * ci_tiles))
     split = min(split, 64, max(1, M // 128))
     max_split_mem = (128 << 20) // (Co * Ci * 4)
     split = max(1, min(split, max_split_mem))

@@ -443,7 +443,7 @@ void conv3x3_nhwc_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor y,
   // shape 239 -> 176 TF with split, M*Co = 3.2M; 7x7 shape 54 -> 187
   // TF, M*Co = 0.8M)
   const bool want_split = tiles < 384 && M * Co <= (2ll << 20);
-  int64_t split = want_split ? 768 / (tiles > 0 ? tiles : 1) : 1;
+  int64_t split = want_split ? 512 / (tiles > 0 ? tiles : 1) : 1;
   if (split > KT / 2) split = KT / 2;
   const int64_t max_mem_split = (64ll << 20) / (M * Co * 4);
   if (split > max_mem_split) split = max_mem_split;
@@ -487,7 +487,7 @@ void conv3x3_wgrad_bf16(torch::Tensor x, torch::Tensor dy,
   const int64_t M = (int64_t)Nb * Ho * Wo;
   const int K9 = 9 * Ci;
   const int64_t tiles = ((Co + 127) / 128) * (int64_t)((K9 + 127) / 128);
-  int64_t split = 768 / (tiles > 0 ? tiles : 1);
+  int64_t split = 512 / (tiles > 0 ? tiles : 1);
   split = std::min<int64_t>(split, 64);
   split = std::min<int64_t>(split, std::max<int64_t>(1, M / 128));
   const int64_t max_mem = (128ll << 20) / ((int64_t)Co * K9 * 4);
