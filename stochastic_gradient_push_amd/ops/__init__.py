@@ -227,7 +227,7 @@ def gemm_nt(A: torch.Tensor, B: torch.Tensor,
     if M <= 2048:
         split = 0
         if K >= 1024 and tiles <= 256:
-            split = min(64, max(2, 768 // tiles), K // 128,
+            split = min(64, max(2, 512 // tiles), K // 128,
                         (64 << 20) // (M * N * 4))
         if split >= 2:
             ext.gemm_nt_splitk_bf16(A, B, C, split)
