@@ -359,3 +359,62 @@ def test_osgp_graphed_step_order_consensus():
 
 def test_osgp_graphed_step_order_consensus_w4():
     run_dist(_osgp_graphed_order, world_size=4)
+
+
+def _weighted_mixing_wrapper(rank, world_size):
+    """GossipDataParallel with NON-uniform mixing (WeightedMixing):
+    forces the non-lazy bias/de-bias path (lazy_mixing False) and the
+    ps-weight-on-the-wire message format, combined with bf16 working
+    weights (master rescales must refresh the shadow).  Consensus at
+    zero lr."""
+    from stochastic_gradient_push_amd import GossipDataParallel
+    from stochastic_gradient_push_amd.graphs import RingGraph
+    from stochastic_gradient_push_amd.mixing import WeightedMixing
+    from stochastic_gradient_push_amd.ops.fused_sgd import FusedSGD
+
+    model = tiny_model(seed=rank)
+    cast = [p for p in model.parameters() if p.ndim >= 2]
+    keep = [p for p in model.parameters() if p.ndim < 2]
+    target = torch.cat([p.detach().reshape(-1) for p in cast + keep])
+    dist.all_reduce(target)
+    target /= world_size
+
+    graph = RingGraph(rank, world_size)
+    mixing = WeightedMixing(
+        graph, torch.device("cpu"),
+        {(rank + 1) % world_size: 0.4, (rank - 1) % world_size: 0.25},
+    )
+    gdp = GossipDataParallel(
+        model, graph=graph, mixing=mixing, push_sum=True,
+        working_dtype=torch.bfloat16,
+    )
+    assert not gdp.lazy_mixing  # non-uniform => non-lazy state machine
+    opt = FusedSGD(gdp.flatp, lr=0.0)
+    loss_fn = nn.CrossEntropyLoss()
+    gdp.train()
+    x = torch.randn(2, 3, 8, 8)
+    y = torch.randint(0, 10, (2,))
+    for _ in range(40):
+        with torch.autocast(device_type="cpu", dtype=torch.bfloat16):
+            loss = loss_fn(gdp(x), y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        gdp.transfer_params()
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    gdp.unbias()
+    flat = gdp.flatp.flat.detach()
+    assert torch.allclose(flat, target, atol=2e-3), (
+        f"rank {rank}: max err {(flat - target).abs().max()}"
+    )
+    # shadow still tracks the master through the rescale/merge path
+    assert torch.equal(
+        gdp.flatp.shadow,
+        gdp.flatp.flat[: gdp.flatp.n_cast].to(torch.bfloat16),
+    )
+    gdp.shutdown()
+
+
+def test_weighted_mixing_wrapper_consensus():
+    run_dist(_weighted_mixing_wrapper, world_size=2)
