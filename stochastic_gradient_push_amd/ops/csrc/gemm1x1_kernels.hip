@@ -950,22 +950,20 @@ __global__ __launch_bounds__(512) void k_gemm_nt_bf16_v6(
 // write into the v1 kernel's padded [c][m] LDS image, so the MFMA inner
 // loop is identical to the NT kernel's.
 //
-// Staging geometry (two rounds of hardware-measured rewrites): lanes
-// are M-FASTEST (the first c-fastest mapping was 16-way bank-
-// conflicted and per-element guarded — 3-6x below roofline), global
-// reads are one bf16x8 vector per lane, LDS transposed writes land on
-// consecutive dwords per 16-lane group (2-way).  BK=64 m-rows per
-// step + register-staged double buffering: 32 MFMA per wave between
-// barriers (the BK=32 single-buffer version measured wait/busy 12.6).
-#define TSTRIDE 72  // 64 m-cols + 8 pad per [channel][m] LDS row
-
+// Staging geometry (round-2 rewrite after the first hardware numbers
+// came in 3-6x below roofline): lanes are M-FASTEST — 16-lane groups
+// cover 16 consecutive m-rows of one 8-channel column group, so the
+// global read is ONE bf16x8 vector load per lane (rows Co*2 B apart,
+// no per-element guards) and the 8 transposed LDS writes land on 8
+// consecutive dwords per 16-lane group (2-way conflict) instead of the
+// old c-fastest mapping's single bank (16-way).
 __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
     const ushort_t* __restrict__ dy,  // [M, Co]
     const ushort_t* __restrict__ x,   // [M, Ci]
     float* __restrict__ partials,     // [split][Co, Ci] (tile-major ok)
     int64_t M, int Co, int Ci, int split) {
-  __shared__ ushort_t As[2][BM * TSTRIDE];  // [co][m] images (x2)
-  __shared__ ushort_t Bs[2][BN * TSTRIDE];  // [ci][m]
+  __shared__ ushort_t As[2][BM * LDS_STRIDE];  // [co][m] images (x2)
+  __shared__ ushort_t Bs[2][BN * LDS_STRIDE];  // [ci][m]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -995,22 +993,23 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
 #pragma unroll
       for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    // BK=64 m-rows per step: lanes are m-fastest (mrow = tid & 63), and
-    // each thread stages 4 of the 16 8-channel column groups per
-    // operand — 32 MFMA per wave between barriers (2x the BK=32
-    // version), register-staged double-buffered.
-    const int mrow = tid & 63;
-    const int cgrp = tid >> 6;  // 0..3
+    // seg -> (m-row, channel-group), m-fastest within 16-lane groups
+    const int mrow = tid & 31;         // 32 m-rows per BK step
+    const int cgrp = tid >> 5;         // 8 channel-groups per 256 thr
     const bool a_full = tco + BM <= Co;
     const bool b_full = tci + BN <= Ci;
-    bf16x8 ra[4], rb[4];
+
+    // register-staged double buffering (the NT v2 structure): loads
+    // for m-chunk t+1 issue right after the barrier and hide behind
+    // the MFMA block; writes land in the other LDS buffer
+    bf16x8 ra[2], rb[2];
 
 #define TN_LOAD(k0_)                                                       \
   do {                                                                     \
     const int64_t gm = (k0_) + mrow;                                       \
     const bool mok = gm < m1;                                              \
-    _Pragma("unroll") for (int half = 0; half < 4; ++half) {               \
-      const int c8 = (cgrp + half * 4) * 8;                                \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
       {                                                                    \
         const int gc0 = tco + c8;                                          \
         if (mok && a_full) {                                               \
@@ -1040,42 +1039,39 @@ __global__ __launch_bounds__(256) void k_gemm_tn_partial_bf16(
 
 #define TN_WRITE(buf)                                                      \
   do {                                                                     \
-    _Pragma("unroll") for (int half = 0; half < 4; ++half) {               \
-      const int c8 = (cgrp + half * 4) * 8;                                \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {               \
+      const int c8 = (cgrp + half * 8) * 8;                                \
       const ushort_t* ea = reinterpret_cast<const ushort_t*>(&ra[half]);   \
       const ushort_t* eb = reinterpret_cast<const ushort_t*>(&rb[half]);   \
       _Pragma("unroll") for (int j = 0; j < 8; ++j) {                      \
-        As[buf][(c8 + j) * TSTRIDE + mrow] = ea[j];                        \
-        Bs[buf][(c8 + j) * TSTRIDE + mrow] = eb[j];                        \
+        As[buf][(c8 + j) * LDS_STRIDE + mrow] = ea[j];                     \
+        Bs[buf][(c8 + j) * LDS_STRIDE + mrow] = eb[j];                     \
       }                                                                    \
     }                                                                      \
   } while (0)
 
-    const int KTm = (int)((m1 - m0 + 63) / 64);
+    const int KTm = (int)((m1 - m0 + BK - 1) / BK);
     TN_LOAD(m0);
     TN_WRITE(0);
     for (int kt = 0; kt < KTm; ++kt) {
       __syncthreads();
       const int buf = kt & 1;
-      if (kt + 1 < KTm) TN_LOAD(m0 + (int64_t)(kt + 1) * 64);
+      if (kt + 1 < KTm) TN_LOAD(m0 + (int64_t)(kt + 1) * BK);
 
+      bf16x8 afrag[4], bfrag[4];
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 afrag[4], bfrag[4];
-#pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          afrag[i] = *reinterpret_cast<const bf16x8*>(
-              As[buf] + (wm + i * 16 + frow) * TSTRIDE + ks * 32 + fk0);
-          bfrag[i] = *reinterpret_cast<const bf16x8*>(
-              Bs[buf] + (wn + i * 16 + frow) * TSTRIDE + ks * 32 + fk0);
-        }
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int j = 0; j < 4; ++j)
-            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            As[buf] + (wm + i * 16 + frow) * LDS_STRIDE + fk0);
+        bfrag[i] = *reinterpret_cast<const bf16x8*>(
+            Bs[buf] + (wn + i * 16 + frow) * LDS_STRIDE + fk0);
       }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
 
       __syncthreads();
       if (kt + 1 < KTm) TN_WRITE(buf ^ 1);
