@@ -225,6 +225,21 @@ class FlatParams:
                     ptrs.append(0)
                     fixups.add(id(p))
             if ptrs != grp["ptrs"]:
+                capturing = out.is_cuda and \
+                    torch.cuda.is_current_stream_capturing()
+                if grp.get("captured") and not capturing:
+                    # a hipGraph replay re-executes the captured H2D
+                    # table copy, re-reading THIS host buffer — an
+                    # eager step must never mutate it, so it gets a
+                    # fresh host+device table pair of its own
+                    host = torch.zeros(len(params), dtype=torch.int64)
+                    if out.is_cuda:
+                        host = host.pin_memory()
+                    grp["host"] = host
+                    grp["table"] = torch.zeros(
+                        len(params), dtype=torch.int64, device=out.device
+                    )
+                grp["captured"] = capturing
                 grp["host"].copy_(torch.tensor(ptrs, dtype=torch.int64))
                 grp["table"].copy_(grp["host"], non_blocking=True)
                 grp["ptrs"] = ptrs
