@@ -130,8 +130,11 @@ class _MfmaConv1x1Fn(torch.autograd.Function):
 
 class MfmaConv1x1(nn.Conv2d):
     """1x1 convolution on the hand-written MFMA GEMM kernels (north-star
-    conv path; behind ``conv_impl='mfma'`` — MIOpen remains the default
-    while the kernels trail hipBLASLt, see profiles/r01_summary.md)."""
+    conv path; ``conv_impl='mfma'``/'auto').  The NT/TN ladder beats
+    hipBLASLt on several ResNet shapes (profiles/r02_summary.md);
+    MIOpen stays the measured end-to-end default because its in-graph
+    wgrad/fwd mix is still faster in aggregate (r02 steady-state
+    analysis)."""
 
     def __init__(self, cin, cout, stride=1):
         super().__init__(cin, cout, 1, stride=stride, bias=False)
