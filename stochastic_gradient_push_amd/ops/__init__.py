@@ -72,6 +72,13 @@ def _ext_for(t: torch.Tensor):
 
 def _as_scalar_tensor(a: Scalar, like: torch.Tensor) -> torch.Tensor:
     if isinstance(a, torch.Tensor):
+        # cpu-comm mode keeps push-sum scalars on the comm device (CPU)
+        # while the flat params are CUDA — move on mismatch (cold path;
+        # device-comm mode passes device-resident scalars straight through)
+        if a.device != like.device:
+            a = a.to(like.device)
+        if a.dtype != torch.float32:
+            a = a.float()
         return a
     return torch.tensor([float(a)], device=like.device, dtype=torch.float32)
 
