@@ -167,3 +167,25 @@ def test_experiment_matrix_and_plot(tmp_path):
         results_dir=str(tmp_path / "results_dir"), metric="avg:Loss",
     )
     assert os.path.exists(out)
+
+
+def test_example_distributed_averaging(tmp_path):
+    """The standalone averaging example runs end to end (4 gloo ranks,
+    reference README.md:67-68 usable-standalone contract)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    r = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "4",
+            "--master-addr", "127.0.0.1", "--master-port", "29531",
+            os.path.join(repo, "examples", "distributed_averaging.py"),
+        ],
+        env=env, timeout=300, cwd=str(tmp_path),
+        capture_output=True, text=True,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "converged" in r.stdout.lower() or "average" in r.stdout.lower()
