@@ -189,3 +189,21 @@ def test_example_distributed_averaging(tmp_path):
     )
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     assert "converged" in r.stdout.lower() or "average" in r.stdout.lower()
+
+
+def test_tools_compile():
+    """Every tools/ script parses and the launch scripts are present
+    (keeps utility drift out of the GPU-validated paths)."""
+    import py_compile
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tools = os.path.join(repo, "tools")
+    names = sorted(os.listdir(tools))
+    assert {"analyze_ktrace.py", "bringup_multigpu.py", "conv_bench.py",
+            "gemm_bench.py", "kernels_pmc.py"} <= set(names)
+    for n in names:
+        if n.endswith(".py"):
+            py_compile.compile(os.path.join(tools, n), doraise=True)
+    scripts = os.listdir(os.path.join(repo, "job_scripts"))
+    assert any(s.startswith("submit_SGP") for s in scripts)
+    assert any("single_node" in s for s in scripts)
