@@ -402,6 +402,9 @@ class PushPull(Gossiper):
         send = self._prep_out_msg(out_msg, ps_weight, residual)
         self._exchange(send, residual)
         self.refresh_peers_()
+        # keep mixing_weights['lo'] describing the NEW peer set (same
+        # dynamic-graph consistency as PushSum.mix)
+        self.refresh_mixing_weights_(residual)
         self.clean_msg_buffers_()
         return self.parse_in_msg_buffer(residual)
 
