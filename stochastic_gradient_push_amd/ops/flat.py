@@ -158,12 +158,6 @@ class FlatParams:
     def numel(self) -> int:
         return self.flat.numel()
 
-    def _grad_view(self, offset: int, n: int):
-        if self.shadow is not None and offset < self.n_cast:
-            return self.flat_grad_w.narrow(0, offset, n)
-        base = offset - self.n_cast
-        return self.flat_grad.narrow(0, base, n)
-
     # -- steal-mode gradients ------------------------------------------------
     # With p.grad = None, autograd ASSIGNS each produced gradient tensor
     # (no accumulate-add kernel per parameter).  gather_grads() then
