@@ -115,6 +115,15 @@ def build_parser():
                    choices=["auto", "miopen", "gemm", "mfma"],
                    help="conv backend (auto: hand-written MFMA kernels on "
                         "the measured winning shapes, MIOpen elsewhere)")
+    p.add_argument("--master_weights", default="auto", type=str,
+                   choices=["auto", "on", "off"],
+                   help="bf16 working weights over an fp32 master "
+                        "(auto: on for cuda + fused_sgd)")
+    p.add_argument("--steal_grads", default="auto", type=str,
+                   choices=["auto", "on", "off"],
+                   help="steal-mode grads + fused multi-gather instead of "
+                        "per-param accumulate adds (auto: on for cuda + "
+                        "fused_sgd)")
     p.add_argument("--num_classes", default=1000, type=int)
     p.add_argument("--fused_sgd", default="True", type=str,
                    help="use the one-kernel flat-buffer SGD")
@@ -598,11 +607,23 @@ def main(argv=None):
             use_streams=not args.no_cuda_streams,
             rank=args.rank,
             world_size=args.world_size,
+            working_dtype=(
+                torch.bfloat16 if args.fused_sgd and (
+                    args.master_weights == "on"
+                    or (args.master_weights == "auto"
+                        and args.device == "cuda")
+                ) else None
+            ),
         )
         if args.fused_sgd:
             optimizer = FusedSGD(
                 model.flatp, lr=args.lr, momentum=args.momentum,
                 weight_decay=args.weight_decay, nesterov=args.nesterov,
+                steal_grads=(
+                    args.steal_grads == "on"
+                    or (args.steal_grads == "auto"
+                        and args.device == "cuda")
+                ),
             )
         else:
             optimizer = torch.optim.SGD(
