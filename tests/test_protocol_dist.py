@@ -233,3 +233,48 @@ def _chaos_training(rank, world_size):
 
 def test_chaos_gossip_failures():
     run_dist(_chaos_training, world_size=2)
+
+
+def _asynch_mode(rank, world_size):
+    """synch_freq > 0 (reference distributed.py:336-387): the forward
+    hook polls gossip NON-blocking for up to synch_freq iterations
+    before forcing a blocking wait; asynch mode disables lazy mixing
+    (explicit bias/de-bias) and must still reach consensus at zero lr."""
+    import torch
+    import torch.distributed as dist
+    import torch.nn as nn
+
+    from stochastic_gradient_push_amd import GossipDataParallel
+
+    torch.manual_seed(rank)
+    model = nn.Sequential(nn.Linear(6, 12), nn.ReLU(), nn.Linear(12, 3))
+    target = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    dist.all_reduce(target)
+    target /= world_size
+
+    gdp = GossipDataParallel(model, push_sum=True, synch_freq=3)
+    assert gdp.asynch and not gdp.lazy_mixing
+    opt = torch.optim.SGD(gdp.parameters(), lr=0.0)
+    x = torch.randn(4, 6)
+    y = torch.randn(4, 3)
+    gdp.train()
+    for _ in range(60):
+        loss = ((gdp(x) - y) ** 2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        gdp.transfer_params()
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    gdp.unbias()
+    flat = gdp.flatp.flat.detach()
+    assert torch.allclose(flat, target, atol=2e-3), (
+        f"rank {rank}: max err {(flat - target).abs().max()}"
+    )
+    gdp.shutdown()
+
+
+def test_asynch_synch_freq_consensus():
+    from tests.dist_utils import run_dist
+
+    run_dist(_asynch_mode, world_size=2)
