@@ -418,3 +418,41 @@ def _weighted_mixing_wrapper(rank, world_size):
 
 def test_weighted_mixing_wrapper_consensus():
     run_dist(_weighted_mixing_wrapper, world_size=2)
+
+
+def _eval_drains_gossip(rank, world_size):
+    """model.eval() disables gossip and drains the in-flight exchange so
+    the de-biased estimate includes received residuals (reference
+    distributed.py:322-327); train() re-enables."""
+    from stochastic_gradient_push_amd import GossipDataParallel
+
+    model = tiny_model(seed=rank)
+    gdp = GossipDataParallel(model, push_sum=True)
+    gdp.train()
+    assert gdp.gossip_enable
+    x = torch.randn(2, 3, 8, 8)
+    y = torch.randint(0, 10, (2,))
+    loss_fn = nn.CrossEntropyLoss()
+    loss = loss_fn(gdp(x), y)
+    loss.backward()
+    gdp.transfer_params()  # kick an exchange
+    gdp.eval()             # must drain it and disable gossip
+    assert not gdp.gossip_enable
+    assert not gdp.gossiping
+    with torch.no_grad():
+        out = gdp(x)       # eval forward runs without gossip hooks
+    assert torch.isfinite(out).all()
+    gdp.train()
+    assert gdp.gossip_enable
+    # a fresh round still works after the eval round-trip
+    loss = loss_fn(gdp(x), y)
+    loss.backward()
+    gdp.transfer_params()
+    gdp.sync_comms()
+    gdp._query_gossip_queue(non_blocking=False)
+    assert torch.isfinite(gdp.flatp.flat).all()
+    gdp.shutdown()
+
+
+def test_eval_drains_gossip():
+    run_dist(_eval_drains_gossip, world_size=2)
