@@ -411,12 +411,18 @@ class GossipDataParallel(Module):
         return self
 
     def eval(self):
+        # drain the in-flight exchange FIRST, while gossip is still
+        # enabled: the reference disables gossip and THEN queries
+        # (reference distributed.py:322-327 + the gossip_enable guard at
+        # :338), so its eval-time drain silently never runs and the
+        # de-biased eval estimate misses received residuals — a latent
+        # reference quirk we fix rather than reproduce
+        if self.distributed and self.gossip_enable:
+            self._query_gossip_queue(non_blocking=self.asynch)
         super().eval()
         for rep in self._module_copies[1:]:
             rep.eval()
         self.gossip_enable = False
-        if self.distributed:
-            self._query_gossip_queue(non_blocking=self.asynch)
         return self
 
     def block(self):
