@@ -207,3 +207,36 @@ def test_tools_compile():
     scripts = os.listdir(os.path.join(repo, "job_scripts"))
     assert any(s.startswith("submit_SGP") for s in scripts)
     assert any("single_node" in s for s in scripts)
+
+
+def test_nic_discovery_and_env_pinning(monkeypatch):
+    """NIC autodiscovery + RCCL/Gloo env pinning (reference
+    experiment_utils/helpers.py:44-67, gossip_sgd.py:654-666)."""
+    from stochastic_gradient_push_amd.utils import nic
+
+    # deterministic fake interface universe
+    monkeypatch.setattr(
+        nic.os, "listdir", lambda _: ["lo", "ib0", "ens3f0", "docker0"]
+    )
+
+    class FakeOut:
+        stdout = b"2: ens3f0: <UP> ...\n3: ib0: <UP> ...\n"
+
+    monkeypatch.setattr(nic.subprocess, "run",
+                        lambda *a, **k: FakeOut())
+    assert nic.get_tcp_interface_name("ethernet") == "ens3f0"
+    assert nic.get_tcp_interface_name("infiniband") == "ib0"
+
+    monkeypatch.delenv("GLOO_SOCKET_IFNAME", raising=False)
+    nic.pin_comm_env("gloo", "ethernet")
+    assert nic.os.environ["GLOO_SOCKET_IFNAME"] == "ens3f0"
+    monkeypatch.delenv("NCCL_SOCKET_IFNAME", raising=False)
+    monkeypatch.delenv("NCCL_IB_DISABLE", raising=False)
+    nic.pin_comm_env("nccl", "ethernet")
+    assert nic.os.environ["NCCL_SOCKET_IFNAME"] == "ens3f0"
+    assert nic.os.environ["NCCL_IB_DISABLE"] == "1"
+
+    # interface of the requested type absent -> loud failure
+    monkeypatch.setattr(nic.os, "listdir", lambda _: ["lo", "docker0"])
+    with pytest.raises(RuntimeError):
+        nic.get_tcp_interface_name("infiniband")
