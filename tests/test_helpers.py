@@ -227,14 +227,19 @@ def test_nic_discovery_and_env_pinning(monkeypatch):
     assert nic.get_tcp_interface_name("ethernet") == "ens3f0"
     assert nic.get_tcp_interface_name("infiniband") == "ib0"
 
-    monkeypatch.delenv("GLOO_SOCKET_IFNAME", raising=False)
-    nic.pin_comm_env("gloo", "ethernet")
-    assert nic.os.environ["GLOO_SOCKET_IFNAME"] == "ens3f0"
-    monkeypatch.delenv("NCCL_SOCKET_IFNAME", raising=False)
-    monkeypatch.delenv("NCCL_IB_DISABLE", raising=False)
-    nic.pin_comm_env("nccl", "ethernet")
-    assert nic.os.environ["NCCL_SOCKET_IFNAME"] == "ens3f0"
-    assert nic.os.environ["NCCL_IB_DISABLE"] == "1"
+    # pin_comm_env mutates the real process env; the fake interface
+    # name must NOT leak into later tests (gloo would try to bind to
+    # it), so pop everything this block touches
+    try:
+        nic.pin_comm_env("gloo", "ethernet")
+        assert nic.os.environ["GLOO_SOCKET_IFNAME"] == "ens3f0"
+        nic.pin_comm_env("nccl", "ethernet")
+        assert nic.os.environ["NCCL_SOCKET_IFNAME"] == "ens3f0"
+        assert nic.os.environ["NCCL_IB_DISABLE"] == "1"
+    finally:
+        for var in ("GLOO_SOCKET_IFNAME", "NCCL_SOCKET_IFNAME",
+                    "NCCL_IB_DISABLE"):
+            nic.os.environ.pop(var, None)
 
     # interface of the requested type absent -> loud failure
     monkeypatch.setattr(nic.os, "listdir", lambda _: ["lo", "docker0"])

@@ -116,13 +116,25 @@ def _run_trainer(tmp, extra, script="gossip_sgd.py", nprocs=1, timeout=240):
                        cwd=str(tmp))
     else:
         procs = []
-        for r in range(nprocs):
-            e = dict(env)
-            e["RANK"] = str(r)
-            e["WORLD_SIZE"] = str(nprocs)
-            procs.append(subprocess.Popen(base, env=e, cwd=str(tmp)))
-        for p in procs:
-            assert p.wait(timeout=timeout) == 0
+        try:
+            for r in range(nprocs):
+                e = dict(env)
+                e["RANK"] = str(r)
+                e["WORLD_SIZE"] = str(nprocs)
+                procs.append(subprocess.Popen(base, env=e, cwd=str(tmp)))
+            for p in procs:
+                assert p.wait(timeout=timeout) == 0
+        finally:
+            # never leak rank processes on failure: a survivor keeps
+            # the rendezvous port bound and cascades into later tests
+            for p in procs:
+                if p.poll() is None:
+                    p.kill()
+            for p in procs:
+                try:
+                    p.wait(timeout=10)
+                except Exception:
+                    pass
 
 
 def test_trainer_cli_single_process(tmp_path):
